@@ -1,0 +1,190 @@
+"""Functional masked ops with reference-exact semantics.
+
+The observable contract (reference: utils/mask_layers.py:23-34,59-70,
+104-119): forward computes with ``mask * weight``; the autograd chain
+therefore multiplies the weight gradient by the mask. Raw (unmasked)
+weights keep receiving weight-decay/momentum updates — they are nullified
+at forward time only — and checkpoints contain the raw weights.
+
+MI355X design: ``mask_apply`` is a single fused HIP kernel producing the
+compute-dtype masked weight (no fp32 temp + separate autocast cast), and
+layers may hold a *cached* masked weight maintained by the fused SGD
+kernel so steady-state forwards skip the multiply entirely.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from turboprune_amd.ops import _backend
+
+
+def mask_apply(weight: torch.Tensor, mask: torch.Tensor,
+               out_dtype: Optional[torch.dtype] = None) -> torch.Tensor:
+    """out = (mask != 0) ? weight : 0, cast to out_dtype. No autograd."""
+    out_dtype = out_dtype or weight.dtype
+    if _backend.use_native(weight, mask):
+        return _backend.extension().mask_apply(weight, mask, out_dtype)
+    return (weight * mask).to(out_dtype)
+
+
+def grad_mask_apply(grad: torch.Tensor, mask: torch.Tensor,
+                    out_dtype: torch.dtype) -> torch.Tensor:
+    """grad_weight = mask ⊙ grad, cast to the raw weight dtype."""
+    if _backend.use_native(grad, mask):
+        return _backend.extension().mask_apply(grad, mask, out_dtype)
+    return (grad * mask.to(grad.dtype)).to(out_dtype)
+
+
+class MaskedWeight(torch.autograd.Function):
+    """Differentiable masked weight: forward yields ``mask ⊙ weight`` (or a
+    pre-computed cache of it), backward applies the mask to the gradient and
+    casts to the raw weight's dtype."""
+
+    @staticmethod
+    def forward(ctx, weight: torch.Tensor, mask: torch.Tensor,
+                cache: Optional[torch.Tensor], compute_dtype: Optional[torch.dtype]):
+        ctx.save_for_backward(mask)
+        ctx.weight_dtype = weight.dtype
+        if cache is not None:
+            return cache
+        return mask_apply(weight, mask, compute_dtype or weight.dtype)
+
+    @staticmethod
+    def backward(ctx, grad):
+        (mask,) = ctx.saved_tensors
+        return grad_mask_apply(grad, mask, ctx.weight_dtype), None, None, None
+
+
+def masked_weight(weight: torch.Tensor, mask: torch.Tensor,
+                  cache: Optional[torch.Tensor] = None,
+                  compute_dtype: Optional[torch.dtype] = None) -> torch.Tensor:
+    return MaskedWeight.apply(weight, mask, cache, compute_dtype)
+
+
+def masked_conv2d(x, weight, mask, bias=None, stride=1, padding=0,
+                  dilation=1, groups=1, cache=None, compute_dtype=None):
+    w = masked_weight(weight, mask, cache, compute_dtype)
+    return F.conv2d(x, w, bias, stride, padding, dilation, groups)
+
+
+def masked_linear(x, weight, mask, bias=None, cache=None, compute_dtype=None):
+    w = masked_weight(weight, mask, cache, compute_dtype)
+    if _backend.use_native(x, w) and x.dim() >= 2:
+        ext = _backend.extension()
+        if ext is not None and hasattr(ext, "masked_linear_available") \
+                and ext.masked_linear_available(x, w):
+            return _MaskedLinearGemm.apply(x, w, bias)
+    return F.linear(x, w, bias)
+
+
+class _MaskedLinearGemm(torch.autograd.Function):
+    """Linear on the in-house MFMA GEMM kernel (y = x @ w^T + b).
+
+    Forward/backward both route through the HIP GEMM; backward grads:
+      grad_x = grad_y @ w ; grad_w = grad_y^T @ x ; grad_b = sum(grad_y).
+    """
+
+    @staticmethod
+    def forward(ctx, x, w, bias):
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = bias is not None
+        ext = _backend.extension()
+        return ext.linear_fwd(x, w, bias)
+
+    @staticmethod
+    def backward(ctx, grad_y):
+        x, w = ctx.saved_tensors
+        ext = _backend.extension()
+        grad_x, grad_w = ext.linear_bwd(grad_y.contiguous(), x, w)
+        grad_b = None
+        if ctx.has_bias:
+            grad_b = grad_y.reshape(-1, grad_y.shape[-1]).sum(0)
+        return grad_x, grad_w, grad_b
+
+
+def bernoulli_mask_(mask: torch.Tensor, p: float,
+                    seed: Optional[int] = None) -> torch.Tensor:
+    """In-place Bernoulli(p) fill of a mask buffer (reference:
+    mask_layers.py:43 ``zeros_like(weight).bernoulli_(p)``).
+
+    On GPU a Philox counter-based HIP kernel is used (seeded from torch's
+    RNG unless an explicit seed is given)."""
+    if _backend.use_native(mask):
+        if seed is None:
+            seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
+        _backend.extension().bernoulli_mask_(mask, float(p), seed)
+        return mask
+    return mask.zero_().bernoulli_(p)
+
+
+def kth_smallest(values: torch.Tensor, k: int) -> float:
+    """k-th smallest (1-based) of a 1-D tensor — the pruning threshold
+    primitive (reference: torch.kthvalue at utils/pruning_utils.py:79).
+
+    GPU path: multi-pass radix select HIP kernel (no full sort)."""
+    assert values.dim() == 1
+    assert 1 <= k <= values.numel(), (k, values.numel())
+    if _backend.use_native(values):
+        return float(_backend.extension().kth_smallest(values, k))
+    return float(torch.kthvalue(values.float().cpu(), k).values)
+
+
+def mask_from_threshold_(mask: torch.Tensor, score: torch.Tensor,
+                         threshold: float) -> torch.Tensor:
+    """mask = where(score <= threshold, 0, 1) in place (reference:
+    utils/pruning_utils.py:82-88)."""
+    if _backend.use_native(mask, score):
+        _backend.extension().mask_from_threshold_(mask, score, float(threshold))
+        return mask
+    mask.copy_(torch.where(score <= threshold,
+                           torch.zeros_like(mask), torch.ones_like(mask)))
+    return mask
+
+
+def masked_abs_score(weight: torch.Tensor, mask: torch.Tensor,
+                     other: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """|mask * weight| (magnitude) or |mask * weight * other| (SNIP/SynFlow
+    style with other=grad). Reference: utils/pruning_utils.py:75,189-191."""
+    if _backend.use_native(weight, mask):
+        return _backend.extension().masked_abs_score(
+            weight, mask, other if other is not None else torch.Tensor())
+    s = (mask * weight).detach()
+    if other is not None:
+        s = s * other
+    return s.abs()
+
+
+def cross_entropy(logits: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    """Mean-reduction cross-entropy. GPU: fused log-softmax+NLL fwd/bwd HIP
+    kernel (one pass over logits each way); CPU: torch reference."""
+    if _backend.use_native(logits) and logits.dim() == 2:
+        return _FusedCrossEntropy.apply(logits.contiguous(), target)
+    return F.cross_entropy(logits, target)
+
+
+class _FusedCrossEntropy(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, target):
+        ext = _backend.extension()
+        loss, lse = ext.ce_fwd(logits, target)
+        ctx.save_for_backward(logits, target, lse)
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        logits, target, lse = ctx.saved_tensors
+        ext = _backend.extension()
+        grad_logits = ext.ce_bwd(logits, target, lse, grad_out)
+        return grad_logits, None
+
+
+def accuracy_count(logits: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    """Number of argmax(logits)==target hits, as a device tensor (reference:
+    torchmetrics Accuracy, base_harness.py:54-60 — synced per epoch here)."""
+    if _backend.use_native(logits) and logits.dim() == 2:
+        return _backend.extension().accuracy_count(logits.contiguous(), target)
+    return (logits.argmax(dim=-1) == target).sum()

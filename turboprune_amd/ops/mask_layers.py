@@ -1,0 +1,147 @@
+"""Masked layers — the sparsity mechanism.
+
+Surface-compatible with the reference (utils/mask_layers.py): classes
+``ConvMask`` / ``LinearMask`` / ``Conv1dMask`` with an fp32 ``mask`` buffer
+shaped like the weight (it rides in ``state_dict`` under ``...mask`` keys)
+and a ``set_er_mask(p)`` Bernoulli initializer.
+
+MI355X-first differences (observable behavior unchanged):
+
+- Forward uses a *cached masked compute weight* when the fused optimizer
+  maintains one (``weight_masked``, non-persistent, compute dtype): in
+  steady state the per-forward ``mask*weight`` multiply (reference
+  mask_layers.py:25 — a full weight-sized temp per forward) disappears;
+  the fused SGD kernel rewrites the cache in the same pass as the update.
+- Without a cache, the multiply runs as one fused HIP kernel producing the
+  compute dtype directly.
+- ``Conv1dMask`` keeps the reference's weight shape ``(out, in, 1)`` for
+  checkpoint compatibility but evaluates as a masked linear (kernel-size-1
+  conv ≡ GEMM), and also accepts (B, N, C) token inputs — fixing the
+  reference's latent-broken DeiT path.
+
+Cache contract: whoever mutates ``weight`` or ``mask`` outside the fused
+optimizer must call ``refresh_cache()`` (``load_state_dict`` does this
+automatically); the fused SGD step keeps the cache valid itself.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn as nn
+
+from turboprune_amd.ops import functional as TF
+
+
+class _MaskedMixin:
+    """Shared mask/cache machinery for the three masked layer types."""
+
+    def _init_mask(self) -> None:
+        self.register_buffer("mask", torch.ones_like(self.weight))
+        # compute-dtype cache maintained by the fused optimizer; never saved,
+        # never auto-moved (recreated on device by enable_cache/refresh_cache)
+        self.weight_masked: Optional[torch.Tensor] = None
+        self.compute_dtype: Optional[torch.dtype] = None
+
+    def set_er_mask(self, p: float, seed: Optional[int] = None) -> None:
+        TF.bernoulli_mask_(self.mask, float(p), seed)
+        self.refresh_cache()
+
+    # --- masked-weight cache ---------------------------------------------
+    def enable_cache(self, compute_dtype: torch.dtype) -> None:
+        self.compute_dtype = compute_dtype
+        self.weight_masked = None
+        self.refresh_cache()
+
+    def disable_cache(self) -> None:
+        self.weight_masked = None
+        self.compute_dtype = None
+
+    def refresh_cache(self) -> None:
+        if self.compute_dtype is None:
+            return
+        with torch.no_grad():
+            wm = TF.mask_apply(self.weight, self.mask, self.compute_dtype)
+        if self.weight_masked is not None and \
+                self.weight_masked.shape == wm.shape and \
+                self.weight_masked.dtype == wm.dtype and \
+                self.weight_masked.device == wm.device:
+            self.weight_masked.copy_(wm)
+        else:
+            self.weight_masked = wm
+
+    def _load_from_state_dict(self, *args, **kwargs):
+        super()._load_from_state_dict(*args, **kwargs)
+        self.refresh_cache()
+
+    # sparsity accounting
+    @torch.no_grad()
+    def sparsity(self) -> float:
+        return float((self.mask == 0).sum().item()) / self.mask.numel()
+
+
+def _bias_like(bias: Optional[torch.Tensor], w: torch.Tensor):
+    if bias is not None and bias.dtype != w.dtype \
+            and not torch.is_autocast_enabled():
+        return bias.to(w.dtype)
+    return bias
+
+
+class ConvMask(_MaskedMixin, nn.Conv2d):
+    """Conv2d with a multiplicative 0/1 weight mask (reference:
+    utils/mask_layers.py:10-43)."""
+
+    def __init__(self, **kwargs) -> None:
+        super().__init__(**kwargs)
+        self._init_mask()
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        w = TF.masked_weight(self.weight, self.mask, self.weight_masked,
+                             self.compute_dtype)
+        return torch.nn.functional.conv2d(
+            x, w, _bias_like(self.bias, w), self.stride, self.padding,
+            self.dilation, self.groups)
+
+
+class LinearMask(_MaskedMixin, nn.Linear):
+    """Linear with a multiplicative 0/1 weight mask (reference:
+    utils/mask_layers.py:46-79)."""
+
+    def __init__(self, **kwargs) -> None:
+        super().__init__(**kwargs)
+        self._init_mask()
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return TF.masked_linear(x, self.weight, self.mask, self.bias,
+                                self.weight_masked, self.compute_dtype)
+
+
+class Conv1dMask(_MaskedMixin, nn.Conv1d):
+    """Linear-as-1x1-conv with weight mask (reference:
+    utils/mask_layers.py:82-128). Weight shape (out, in, 1) as in the
+    reference; forward evaluates the equivalent masked linear so (B, C)
+    and (B, N, C) inputs both work."""
+
+    def __init__(self, in_features: int, out_features: int,
+                 bias: bool = False) -> None:
+        super().__init__(in_channels=in_features, out_channels=out_features,
+                         kernel_size=1, stride=1, bias=bias)
+        self._init_mask()
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        cache = self.weight_masked
+        return TF.masked_linear(
+            x, self.weight.squeeze(-1), self.mask.squeeze(-1), self.bias,
+            cache.squeeze(-1) if cache is not None else None,
+            self.compute_dtype)
+
+
+MASKED_LAYER_TYPES = (ConvMask, LinearMask, Conv1dMask)
+
+
+def masked_modules(model: nn.Module):
+    """Iterate (name, module) over all masked layers of a model."""
+    for n, m in model.named_modules():
+        if isinstance(m, MASKED_LAYER_TYPES):
+            yield n, m
