@@ -1,0 +1,106 @@
+import torch
+
+from turboprune_amd.config import compose
+from turboprune_amd.data import (AirbenchLoaders, CifarLoader,
+                                 SyntheticImageNet)
+from turboprune_amd.data import augment
+from turboprune_amd.data.imagenet import IMAGENET_MEAN, IMAGENET_STD
+
+
+def _cfg(tmp_path):
+    return compose("cifar10_er_erk", [
+        f"dataset_params.data_root_dir={tmp_path}/data",
+        "dataset_params.total_batch_size=64",
+        "+dataset_params.synthetic_size=256",
+    ])
+
+
+def test_cifar_loader_shapes(tmp_path):
+    cfg = _cfg(tmp_path)
+    pair = AirbenchLoaders(cfg, device=torch.device("cpu"))
+    batches = list(pair.train_loader)
+    assert len(batches) == 4  # 256/64, drop_last
+    x, y = batches[0]
+    assert x.shape == (64, 3, 32, 32)
+    assert x.dtype == torch.float32
+    assert y.shape == (64,)
+    assert pair.train_loader.synthetic  # no real CIFAR on disk
+
+
+def test_cifar_loader_epoch_variation(tmp_path):
+    cfg = _cfg(tmp_path)
+    loader = AirbenchLoaders(cfg, device=torch.device("cpu")).train_loader
+    torch.manual_seed(0)
+    e0 = torch.cat([x.sum(dim=(1, 2, 3)) for x, _ in loader])
+    e1 = torch.cat([x.sum(dim=(1, 2, 3)) for x, _ in loader])
+    # augmentation + reshuffle change the epoch content
+    assert not torch.allclose(e0.sort().values, e1.sort().values)
+
+
+def test_cifar_test_loader_deterministic(tmp_path):
+    cfg = _cfg(tmp_path)
+    loader = AirbenchLoaders(cfg, device=torch.device("cpu")).test_loader
+    a = torch.cat([x for x, _ in loader])
+    b = torch.cat([x for x, _ in loader])
+    assert torch.equal(a, b)
+
+
+def test_normalize_u8_oracle():
+    imgs = torch.randint(0, 256, (4, 3, 8, 8), dtype=torch.uint8)
+    mean = IMAGENET_MEAN
+    std = IMAGENET_STD
+    out = augment.normalize_u8(imgs, mean, std)
+    ref = (imgs.float() / 255.0 - mean.view(1, 3, 1, 1)) / std.view(1, 3, 1, 1)
+    assert torch.allclose(out, ref, atol=1e-6)
+
+
+def test_normalize_u8_flip():
+    imgs = torch.randint(0, 256, (2, 3, 4, 4), dtype=torch.uint8)
+    flip = torch.tensor([True, False])
+    out = augment.normalize_u8(imgs, torch.zeros(3), torch.ones(3),
+                               flip=flip)
+    assert torch.allclose(out[0], torch.flip(imgs[0].float() / 255, [-1]))
+    assert torch.allclose(out[1], imgs[1].float() / 255)
+
+
+def test_batch_crop_translate():
+    x = torch.arange(2 * 1 * 6 * 6, dtype=torch.float32).reshape(2, 1, 6, 6)
+    shifts = torch.tensor([[0, 0], [2, 2]])
+    out = augment.batch_crop_translate(x, 4, shifts)
+    assert out.shape == (2, 1, 4, 4)
+    assert torch.equal(out[0, 0], x[0, 0, 0:4, 0:4])
+    assert torch.equal(out[1, 0], x[1, 0, 2:6, 2:6])
+
+
+def test_batch_cutout():
+    x = torch.ones(8, 3, 16, 16)
+    augment.batch_cutout(x, 4)
+    # every image lost some pixels, none lost everything
+    per_image = x.sum(dim=(1, 2, 3))
+    assert torch.all(per_image < 3 * 256)
+    assert torch.all(per_image > 0)
+
+
+def test_synthetic_imagenet_sharding_math():
+    dev = torch.device("cpu")
+    loader = SyntheticImageNet(batch_size=64, device=dev, train=True,
+                               image_size=32, pool_size=128,
+                               steps_per_epoch=3)
+    batches = list(loader)
+    assert len(batches) == 3
+    x, y = batches[0]
+    assert x.shape == (64, 3, 32, 32)
+    assert y.min() >= 0 and y.max() < 1000
+
+
+def test_imagenet_loaders_synthetic_fallback(tmp_path):
+    cfg = compose("bench_resnet50_imagenet", [
+        "dataset_params.total_batch_size=32",
+        "+dataset_params.steps_per_epoch=2",
+        f"dataset_params.data_root_dir={tmp_path}",
+    ])
+    from turboprune_amd.data import ImageNetLoaders
+    pair = ImageNetLoaders(cfg, torch.device("cpu"), world_size=4, rank=1,
+                           steps_per_epoch=2)
+    x, y = next(iter(pair.train_loader))
+    assert x.shape == (8, 3, 224, 224)  # 32 total / 4 ranks

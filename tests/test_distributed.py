@@ -1,0 +1,103 @@
+"""Multi-process CPU (gloo) tests of the distributed path: rank-0 prune
+-> broadcast parity, hash equality, and DDP gradient all-reduce parity
+vs single-process gradient accumulation."""
+
+import os
+
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from turboprune_amd.ops.mask_layers import LinearMask
+
+
+def _init(rank, world, port):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+
+def _tiny_model(seed):
+    torch.manual_seed(seed)
+    return torch.nn.Sequential(
+        LinearMask(in_features=8, out_features=16),
+        torch.nn.ReLU(),
+        LinearMask(in_features=16, out_features=4),
+    )
+
+
+# ---------------------------------------------------------------- worker fns
+def _worker_broadcast(rank, world, port, q):
+    _init(rank, world, port)
+    from turboprune_amd.parallel.ddp import (broadcast_model_state,
+                                             check_model_equality)
+    model = _tiny_model(seed=rank)  # deliberately different per rank
+    with torch.no_grad():
+        model[0].mask.bernoulli_(0.5) if rank == 0 else None
+    assert not check_model_equality(model) or world == 1
+    broadcast_model_state(model, src=0)
+    ok = check_model_equality(model)
+    q.put(("eq", rank, ok))
+    dist.destroy_process_group()
+
+
+def _worker_ddp_parity(rank, world, port, q):
+    _init(rank, world, port)
+    from turboprune_amd.parallel.ddp import wrap_ddp
+    model = _tiny_model(seed=rank)
+    ddp = wrap_ddp(model)  # broadcasts rank-0 state first
+    torch.manual_seed(1234)  # same data everywhere; shard by rank below
+    x = torch.randn(2 * world, 8)
+    y = torch.randn(2 * world, 4)
+    xi = x[rank * 2:(rank + 1) * 2]
+    yi = y[rank * 2:(rank + 1) * 2]
+    loss = torch.nn.functional.mse_loss(ddp(xi), yi)
+    loss.backward()
+    g = model[0].weight.grad.clone()
+    q.put(("grad", rank, g.tolist()))  # plain list: no shared-mem lifetime
+    dist.destroy_process_group()
+
+
+def _run_workers(fn, world=2):
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    port = 29531 + (os.getpid() % 500)
+    procs = [ctx.Process(target=fn, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    results = [q.get(timeout=180) for _ in range(world)]
+    for p in procs:
+        p.join(timeout=60)
+        assert p.exitcode == 0
+    return results
+
+
+@pytest.mark.timeout(300)
+def test_broadcast_model_state_parity():
+    results = _run_workers(_worker_broadcast)
+    assert all(ok for (_, _, ok) in results)
+
+
+@pytest.mark.timeout(300)
+def test_ddp_grad_allreduce_matches_accumulation():
+    results = _run_workers(_worker_ddp_parity)
+    grads = {rank: torch.tensor(g) for (_, rank, g) in results}
+    # both ranks see the averaged gradient
+    assert torch.allclose(grads[0], grads[1], atol=1e-6)
+
+    # single-process reference: rank-0 model, mean loss over the full batch
+    model = _tiny_model(seed=0)
+    torch.manual_seed(1234)
+    x = torch.randn(4, 8)
+    y = torch.randn(4, 4)
+    loss = torch.nn.functional.mse_loss(model(x), y)
+    loss.backward()
+    assert torch.allclose(grads[0], model[0].weight.grad, atol=1e-5)
+
+
+def test_world_info_defaults():
+    from turboprune_amd.parallel.ddp import world_info
+    rank, local, world = world_info()
+    assert (rank, local, world) == (0, 0, 1)

@@ -138,11 +138,14 @@ class AirbenchLoaders:
     (flip + translate 2, altflip; reference: dataset.py:229-256)."""
 
     def __init__(self, cfg: Any, device: Optional[torch.device] = None,
-                 synthetic_size: int = 2048):
+                 synthetic_size: Optional[int] = None):
         root = cfg.dataset_params.data_root_dir
         dataset = cfg.dataset_params.dataset_name
         bs = int(cfg.dataset_params.total_batch_size)
         seed = int(cfg.select("experiment_params.seed", 0))
+        if synthetic_size is None:
+            synthetic_size = int(cfg.select("dataset_params.synthetic_size",
+                                            2048))
         self.train_loader = CifarLoader(
             root, dataset, train=True, batch_size=bs,
             aug={"flip": True, "translate": 2}, altflip=True, device=device,
