@@ -1,0 +1,56 @@
+// Python bindings for the turboprune_amd gfx950 kernels.
+#include <torch/extension.h>
+
+namespace turboprune {
+at::Tensor mask_apply(const at::Tensor&, const at::Tensor&, at::ScalarType);
+void mask_from_threshold_(at::Tensor, const at::Tensor&, double);
+at::Tensor masked_abs_score(const at::Tensor&, const at::Tensor&,
+                            const at::Tensor&);
+void bernoulli_mask_(at::Tensor, double, int64_t);
+void sgd_step_(at::Tensor, const at::Tensor&, at::Tensor, const at::Tensor&,
+               at::Tensor, double, double, double);
+double kth_smallest(const at::Tensor&, int64_t);
+std::tuple<at::Tensor, at::Tensor> ce_fwd(const at::Tensor&,
+                                          const at::Tensor&);
+at::Tensor ce_bwd(const at::Tensor&, const at::Tensor&, const at::Tensor&,
+                  const at::Tensor&);
+at::Tensor accuracy_count(const at::Tensor&, const at::Tensor&);
+at::Tensor normalize_u8(const at::Tensor&, const at::Tensor&,
+                        const at::Tensor&, const at::Tensor&, at::ScalarType);
+// masked MFMA GEMM (gemm_masked.hip)
+bool masked_linear_available(const at::Tensor&, const at::Tensor&);
+at::Tensor linear_fwd(const at::Tensor&, const at::Tensor&,
+                      const c10::optional<at::Tensor>&);
+std::tuple<at::Tensor, at::Tensor> linear_bwd(const at::Tensor&,
+                                              const at::Tensor&,
+                                              const at::Tensor&);
+at::Tensor gemm_bf16(const at::Tensor&, const at::Tensor&, bool, bool);
+}  // namespace turboprune
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "turboprune_amd CDNA4 (gfx950) kernels";
+  m.def("mask_apply", &turboprune::mask_apply,
+        "out = (mask!=0 ? weight : 0).to(dtype)");
+  m.def("mask_from_threshold_", &turboprune::mask_from_threshold_,
+        "mask = where(score <= thr, 0, 1) in-place");
+  m.def("masked_abs_score", &turboprune::masked_abs_score,
+        "|w*m| or |w*m*g|");
+  m.def("bernoulli_mask_", &turboprune::bernoulli_mask_,
+        "Philox Bernoulli(p) mask fill");
+  m.def("sgd_step_", &turboprune::sgd_step_,
+        "fused SGD momentum+wd step with mask-reapply cache rewrite");
+  m.def("kth_smallest", &turboprune::kth_smallest,
+        "radix-select k-th smallest of a 1-D fp32 tensor");
+  m.def("ce_fwd", &turboprune::ce_fwd, "fused CE forward -> (loss, lse)");
+  m.def("ce_bwd", &turboprune::ce_bwd, "fused CE backward");
+  m.def("accuracy_count", &turboprune::accuracy_count,
+        "count(argmax(logits)==target)");
+  m.def("normalize_u8", &turboprune::normalize_u8,
+        "fused u8->normalized float/bf16 with optional flip");
+  m.def("masked_linear_available", &turboprune::masked_linear_available);
+  m.def("linear_fwd", &turboprune::linear_fwd, "MFMA GEMM y = x @ w^T + b");
+  m.def("linear_bwd", &turboprune::linear_bwd,
+        "MFMA GEMM grads (grad_x, grad_w)");
+  m.def("gemm_bf16", &turboprune::gemm_bf16,
+        "raw MFMA bf16 GEMM (testing entry)");
+}

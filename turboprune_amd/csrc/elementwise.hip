@@ -1,0 +1,214 @@
+// Elementwise tier: mask application, score construction, mask rewrite,
+// Bernoulli mask generation.
+//
+// All kernels are memory-bound grid-stride loops vectorized to 16 B/lane
+// (guide §6 G13: 64-lane waves, float4 loads); masks/weights are fp32,
+// outputs fp32 or bf16 (the compute dtype).
+
+#include <ATen/ATen.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace turboprune {
+
+// ---- mask_apply: out = (mask != 0) ? w : 0, cast to OutT ----------------
+template <typename OutT>
+__global__ void mask_apply_kernel_v4(const float* __restrict__ w,
+                                     const float* __restrict__ m,
+                                     OutT* __restrict__ out, int64_t n4) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    float4 wv = reinterpret_cast<const float4*>(w)[i];
+    float4 mv = reinterpret_cast<const float4*>(m)[i];
+    OutT o[4];
+    o[0] = from_float<OutT>(mv.x != 0.f ? wv.x : 0.f);
+    o[1] = from_float<OutT>(mv.y != 0.f ? wv.y : 0.f);
+    o[2] = from_float<OutT>(mv.z != 0.f ? wv.z : 0.f);
+    o[3] = from_float<OutT>(mv.w != 0.f ? wv.w : 0.f);
+    *reinterpret_cast<uint4*>(&out[i * 4]) =
+        *reinterpret_cast<const uint4*>(&o[0]);
+  }
+}
+
+template <typename OutT>
+__global__ void mask_apply_kernel(const float* __restrict__ w,
+                                  const float* __restrict__ m,
+                                  OutT* __restrict__ out, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+    out[i] = from_float<OutT>(m[i] != 0.f ? w[i] : 0.f);
+}
+
+// OutT must be 4 bytes for the v4 uint4 store trick to hold 16 B; for
+// bf16 out we store 4x2 B = 8 B via uint2.
+template <>
+__global__ void mask_apply_kernel_v4<__hip_bfloat16>(
+    const float* __restrict__ w, const float* __restrict__ m,
+    __hip_bfloat16* __restrict__ out, int64_t n4) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    float4 wv = reinterpret_cast<const float4*>(w)[i];
+    float4 mv = reinterpret_cast<const float4*>(m)[i];
+    __hip_bfloat16 o[4];
+    o[0] = __float2bfloat16(mv.x != 0.f ? wv.x : 0.f);
+    o[1] = __float2bfloat16(mv.y != 0.f ? wv.y : 0.f);
+    o[2] = __float2bfloat16(mv.z != 0.f ? wv.z : 0.f);
+    o[3] = __float2bfloat16(mv.w != 0.f ? wv.w : 0.f);
+    *reinterpret_cast<uint2*>(&out[i * 4]) =
+        *reinterpret_cast<const uint2*>(&o[0]);
+  }
+}
+
+at::Tensor mask_apply(const at::Tensor& weight, const at::Tensor& mask,
+                      at::ScalarType out_dtype) {
+  TORCH_CHECK(weight.is_cuda() && mask.is_cuda(), "expected GPU tensors");
+  auto w = weight.contiguous().to(at::kFloat);
+  auto m = mask.contiguous().to(at::kFloat);
+  TORCH_CHECK(w.numel() == m.numel(), "weight/mask numel mismatch");
+  auto out = at::empty_like(w, w.options().dtype(out_dtype));
+  int64_t n = w.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  bool vec4 = (n % 4 == 0);
+  if (out_dtype == at::kFloat) {
+    if (vec4) {
+      hipLaunchKernelGGL(mask_apply_kernel_v4<float>,
+                         dim3(elementwise_grid(n / 4)), dim3(kBlock), 0,
+                         stream, w.data_ptr<float>(), m.data_ptr<float>(),
+                         out.data_ptr<float>(), n / 4);
+    } else {
+      hipLaunchKernelGGL(mask_apply_kernel<float>,
+                         dim3(elementwise_grid(n)), dim3(kBlock), 0, stream,
+                         w.data_ptr<float>(), m.data_ptr<float>(),
+                         out.data_ptr<float>(), n);
+    }
+  } else if (out_dtype == at::kBFloat16) {
+    auto* op = reinterpret_cast<__hip_bfloat16*>(out.data_ptr());
+    if (vec4) {
+      hipLaunchKernelGGL(mask_apply_kernel_v4<__hip_bfloat16>,
+                         dim3(elementwise_grid(n / 4)), dim3(kBlock), 0,
+                         stream, w.data_ptr<float>(), m.data_ptr<float>(),
+                         op, n / 4);
+    } else {
+      hipLaunchKernelGGL(mask_apply_kernel<__hip_bfloat16>,
+                         dim3(elementwise_grid(n)), dim3(kBlock), 0, stream,
+                         w.data_ptr<float>(), m.data_ptr<float>(), op, n);
+    }
+  } else {
+    TORCH_CHECK(false, "mask_apply: unsupported out dtype");
+  }
+  return out;
+}
+
+// ---- mask rewrite: mask = (score <= thr) ? 0 : 1 ------------------------
+__global__ void mask_from_threshold_kernel(float* __restrict__ mask,
+                                           const float* __restrict__ score,
+                                           float thr, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride)
+    mask[i] = score[i] <= thr ? 0.f : 1.f;
+}
+
+void mask_from_threshold_(at::Tensor mask, const at::Tensor& score,
+                          double thr) {
+  TORCH_CHECK(mask.is_cuda() && score.is_cuda());
+  TORCH_CHECK(mask.is_contiguous());
+  auto s = score.contiguous().to(at::kFloat);
+  int64_t n = mask.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(mask_from_threshold_kernel, dim3(elementwise_grid(n)),
+                     dim3(kBlock), 0, stream, mask.data_ptr<float>(),
+                     s.data_ptr<float>(), (float)thr, n);
+}
+
+// ---- score construction: |w*m| or |w*m*g| --------------------------------
+__global__ void masked_abs_score_kernel(const float* __restrict__ w,
+                                        const float* __restrict__ m,
+                                        const float* __restrict__ g,
+                                        float* __restrict__ out, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float v = w[i] * m[i];
+    if (g != nullptr) v *= g[i];
+    out[i] = fabsf(v);
+  }
+}
+
+at::Tensor masked_abs_score(const at::Tensor& weight, const at::Tensor& mask,
+                            const at::Tensor& other) {
+  auto w = weight.contiguous().to(at::kFloat);
+  auto m = mask.contiguous().to(at::kFloat);
+  const float* g = nullptr;
+  at::Tensor o;
+  if (other.defined() && other.numel() > 0) {
+    o = other.contiguous().to(at::kFloat);
+    g = o.data_ptr<float>();
+  }
+  auto out = at::empty_like(w);
+  int64_t n = w.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(masked_abs_score_kernel, dim3(elementwise_grid(n)),
+                     dim3(kBlock), 0, stream, w.data_ptr<float>(),
+                     m.data_ptr<float>(), g, out.data_ptr<float>(), n);
+  return out;
+}
+
+// ---- Bernoulli mask fill (Philox, K9) ------------------------------------
+__global__ void bernoulli_mask_kernel(float* __restrict__ mask, float p,
+                                      uint64_t seed, int64_t n4) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n4;
+       i += stride) {
+    Philox4 r = philox4x32(seed, (uint64_t)i);
+    float4 v;
+    v.x = u32_to_uniform(r.x) <= p ? 1.f : 0.f;
+    v.y = u32_to_uniform(r.y) <= p ? 1.f : 0.f;
+    v.z = u32_to_uniform(r.z) <= p ? 1.f : 0.f;
+    v.w = u32_to_uniform(r.w) <= p ? 1.f : 0.f;
+    int64_t base = i * 4;
+    // n4 counts ceil(n/4) quads; guard the tail
+    float* out = mask + base;
+    int64_t remain = (n4 * 4 - base);  // >= 1
+    if (remain >= 4) {
+      *reinterpret_cast<float4*>(out) = v;
+    } else {
+      out[0] = v.x;
+      if (remain > 1) out[1] = v.y;
+      if (remain > 2) out[2] = v.z;
+    }
+  }
+}
+
+__global__ void bernoulli_tail_kernel(float* __restrict__ m, float p,
+                                      uint64_t seed, int64_t base,
+                                      int64_t n) {
+  int64_t i = base + threadIdx.x;
+  if (i < n) {
+    Philox4 r = philox4x32(seed, (uint64_t)(1ull << 62) + i);
+    m[i] = u32_to_uniform(r.x) <= p ? 1.f : 0.f;
+  }
+}
+
+void bernoulli_mask_(at::Tensor mask, double p, int64_t seed) {
+  TORCH_CHECK(mask.is_cuda() && mask.is_contiguous() &&
+              mask.scalar_type() == at::kFloat);
+  int64_t n = mask.numel();
+  // pad to quads; kernel guards tail against n4*4 (== padded length),
+  // so pass exact quad count with tail handling below
+  int64_t n4 = n / 4;
+  auto stream = at::hip::getCurrentHIPStream();
+  if (n4 > 0)
+    hipLaunchKernelGGL(bernoulli_mask_kernel, dim3(elementwise_grid(n4)),
+                       dim3(kBlock), 0, stream, mask.data_ptr<float>(),
+                       (float)p, (uint64_t)seed, n4);
+  int64_t tail = n - n4 * 4;
+  if (tail > 0)
+    hipLaunchKernelGGL(bernoulli_tail_kernel, dim3(1), dim3(4), 0, stream,
+                       mask.data_ptr<float>(), (float)p, (uint64_t)seed,
+                       n4 * 4, n);
+}
+
+}  // namespace turboprune

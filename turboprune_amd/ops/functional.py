@@ -73,6 +73,9 @@ def masked_conv2d(x, weight, mask, bias=None, stride=1, padding=0,
 
 def masked_linear(x, weight, mask, bias=None, cache=None, compute_dtype=None):
     w = masked_weight(weight, mask, cache, compute_dtype)
+    if (w.is_cuda and w.dtype == torch.bfloat16 and x.dtype != w.dtype
+            and torch.is_autocast_enabled()):
+        x = x.to(torch.bfloat16)  # what autocast would do inside F.linear
     if _backend.use_native(x, w) and x.dim() >= 2:
         ext = _backend.extension()
         if ext is not None and hasattr(ext, "masked_linear_available") \
@@ -91,7 +94,7 @@ class _MaskedLinearGemm(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, w, bias):
         ctx.save_for_backward(x, w)
-        ctx.has_bias = bias is not None
+        ctx.bias_dtype = bias.dtype if bias is not None else None
         ext = _backend.extension()
         return ext.linear_fwd(x, w, bias)
 
@@ -101,8 +104,9 @@ class _MaskedLinearGemm(torch.autograd.Function):
         ext = _backend.extension()
         grad_x, grad_w = ext.linear_bwd(grad_y.contiguous(), x, w)
         grad_b = None
-        if ctx.has_bias:
-            grad_b = grad_y.reshape(-1, grad_y.shape[-1]).sum(0)
+        if ctx.bias_dtype is not None:
+            grad_b = grad_y.reshape(-1, grad_y.shape[-1]).sum(0) \
+                .to(ctx.bias_dtype)
         return grad_x, grad_w, grad_b
 
 
