@@ -63,13 +63,28 @@ __global__ void mask_apply_kernel_v4<__hip_bfloat16>(
   }
 }
 
+// re-layout t to match `like`'s strides (dense); no-op when it already does
+static at::Tensor to_layout_of(const at::Tensor& t, const at::Tensor& like) {
+  if (t.strides().vec() == like.strides().vec() &&
+      t.is_non_overlapping_and_dense())
+    return t;
+  auto out = at::empty_strided(like.sizes(), like.strides(), t.options());
+  out.copy_(t);
+  return out;
+}
+
 at::Tensor mask_apply(const at::Tensor& weight, const at::Tensor& mask,
                       at::ScalarType out_dtype) {
   TORCH_CHECK(weight.is_cuda() && mask.is_cuda(), "expected GPU tensors");
-  auto w = weight.contiguous().to(at::kFloat);
-  auto m = mask.contiguous().to(at::kFloat);
+  // preserve the weight's (possibly channels_last) layout: all tensors
+  // iterate flat in storage order with matching strides
+  auto w = weight.is_non_overlapping_and_dense()
+               ? weight.to(at::kFloat)
+               : weight.contiguous().to(at::kFloat);
+  auto m = to_layout_of(mask.to(at::kFloat), w);
   TORCH_CHECK(w.numel() == m.numel(), "weight/mask numel mismatch");
-  auto out = at::empty_like(w, w.options().dtype(out_dtype));
+  auto out = at::empty_strided(w.sizes(), w.strides(),
+                               w.options().dtype(out_dtype));
   int64_t n = w.numel();
   auto stream = at::hip::getCurrentHIPStream();
   bool vec4 = (n % 4 == 0);
@@ -115,8 +130,8 @@ __global__ void mask_from_threshold_kernel(float* __restrict__ mask,
 void mask_from_threshold_(at::Tensor mask, const at::Tensor& score,
                           double thr) {
   TORCH_CHECK(mask.is_cuda() && score.is_cuda());
-  TORCH_CHECK(mask.is_contiguous());
-  auto s = score.contiguous().to(at::kFloat);
+  TORCH_CHECK(mask.is_non_overlapping_and_dense());
+  auto s = to_layout_of(score.to(at::kFloat), mask);
   int64_t n = mask.numel();
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(mask_from_threshold_kernel, dim3(elementwise_grid(n)),
@@ -140,15 +155,17 @@ __global__ void masked_abs_score_kernel(const float* __restrict__ w,
 
 at::Tensor masked_abs_score(const at::Tensor& weight, const at::Tensor& mask,
                             const at::Tensor& other) {
-  auto w = weight.contiguous().to(at::kFloat);
-  auto m = mask.contiguous().to(at::kFloat);
+  auto w = weight.is_non_overlapping_and_dense()
+               ? weight.to(at::kFloat)
+               : weight.contiguous().to(at::kFloat);
+  auto m = to_layout_of(mask.to(at::kFloat), w);
   const float* g = nullptr;
   at::Tensor o;
   if (other.defined() && other.numel() > 0) {
-    o = other.contiguous().to(at::kFloat);
+    o = to_layout_of(other.to(at::kFloat), w);
     g = o.data_ptr<float>();
   }
-  auto out = at::empty_like(w);
+  auto out = at::empty_strided(w.sizes(), w.strides(), w.options());
   int64_t n = w.numel();
   auto stream = at::hip::getCurrentHIPStream();
   hipLaunchKernelGGL(masked_abs_score_kernel, dim3(elementwise_grid(n)),
@@ -194,7 +211,7 @@ __global__ void bernoulli_tail_kernel(float* __restrict__ m, float p,
 }
 
 void bernoulli_mask_(at::Tensor mask, double p, int64_t seed) {
-  TORCH_CHECK(mask.is_cuda() && mask.is_contiguous() &&
+  TORCH_CHECK(mask.is_cuda() && mask.is_non_overlapping_and_dense() &&
               mask.scalar_type() == at::kFloat);
   int64_t n = mask.numel();
   // pad to quads; kernel guards tail against n4*4 (== padded length),

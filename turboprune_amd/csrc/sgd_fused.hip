@@ -80,16 +80,37 @@ static void launch_sgd(at::Tensor& w, const at::Tensor& g, at::Tensor& buf,
 #undef TP_SGD
 }
 
+// all operand tensors iterate flat in the WEIGHT's storage order; any
+// layout (e.g. channels_last) is fine as long as strides match — tensors
+// in another dense layout are re-laid-out once.
+static at::Tensor match_layout(const at::Tensor& t, const at::Tensor& like) {
+  if (t.strides().vec() == like.strides().vec() &&
+      t.is_non_overlapping_and_dense())
+    return t;
+  auto out = at::empty_strided(like.sizes(), like.strides(), t.options());
+  out.copy_(t);
+  return out;
+}
+
 void sgd_step_(at::Tensor w, const at::Tensor& grad, at::Tensor buf,
                const at::Tensor& mask, at::Tensor cache, double lr,
                double momentum, double wd) {
-  TORCH_CHECK(w.is_cuda() && w.is_contiguous() &&
+  TORCH_CHECK(w.is_cuda() && w.is_non_overlapping_and_dense() &&
               w.scalar_type() == at::kFloat,
-              "sgd_step_: weight must be contiguous fp32 on GPU");
-  auto g = grad.contiguous();
+              "sgd_step_: weight must be dense fp32 on GPU");
+  auto g = match_layout(grad, w);
   TORCH_CHECK(g.numel() == w.numel());
-  if (buf.defined() && buf.numel() > 0)
-    TORCH_CHECK(buf.is_contiguous() && buf.scalar_type() == at::kFloat);
+  if (buf.defined() && buf.numel() > 0) {
+    TORCH_CHECK(buf.scalar_type() == at::kFloat);
+    TORCH_CHECK(buf.strides().vec() == w.strides().vec(),
+                "sgd_step_: momentum buffer layout must match weight");
+  }
+  if (mask.defined() && mask.numel() > 0)
+    TORCH_CHECK(mask.strides().vec() == w.strides().vec(),
+                "sgd_step_: mask layout must match weight");
+  if (cache.defined() && cache.numel() > 0)
+    TORCH_CHECK(cache.strides().vec() == w.strides().vec(),
+                "sgd_step_: cache layout must match weight");
   auto stream = at::hip::getCurrentHIPStream();
 
   auto cache_t = cache.defined() && cache.numel() > 0
