@@ -25,6 +25,7 @@ sources = [
     os.path.join(CSRC, "ce_loss.hip"),
     os.path.join(CSRC, "augment.hip"),
     os.path.join(CSRC, "gemm_masked.hip"),
+    os.path.join(CSRC, "batchnorm.hip"),
 ]
 
 setup(

@@ -23,6 +23,7 @@ from typing import List, Optional, Type, Union
 import torch
 import torch.nn as nn
 
+from turboprune_amd.ops.bn import FusedBatchNorm2d, bn_act
 from turboprune_amd.ops.mask_layers import Conv1dMask, ConvMask
 
 
@@ -45,21 +46,21 @@ class BasicBlock(nn.Module):
                  downsample: Optional[nn.Module] = None):
         super().__init__()
         self.conv1 = conv3x3(inplanes, planes, stride)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = FusedBatchNorm2d(planes)
         self.relu = nn.ReLU(inplace=True)
         self.conv2 = conv3x3(planes, planes)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = FusedBatchNorm2d(planes)
         self.downsample = downsample
         self.stride = stride
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.bn2(self.conv2(out))
+        # fused BN+ReLU and BN+add+ReLU epilogues (ops/bn.py)
+        out = bn_act(self.bn1, self.conv1(x), relu=True)
+        out = self.conv2(out)
         if self.downsample is not None:
             identity = self.downsample(x)
-        out += identity
-        return self.relu(out)
+        return bn_act(self.bn2, out, residual=identity, relu=True)
 
 
 class Bottleneck(nn.Module):
@@ -69,24 +70,23 @@ class Bottleneck(nn.Module):
                  downsample: Optional[nn.Module] = None):
         super().__init__()
         self.conv1 = conv1x1(inplanes, planes)
-        self.bn1 = nn.BatchNorm2d(planes)
+        self.bn1 = FusedBatchNorm2d(planes)
         self.conv2 = conv3x3(planes, planes, stride)
-        self.bn2 = nn.BatchNorm2d(planes)
+        self.bn2 = FusedBatchNorm2d(planes)
         self.conv3 = conv1x1(planes, planes * self.expansion)
-        self.bn3 = nn.BatchNorm2d(planes * self.expansion)
+        self.bn3 = FusedBatchNorm2d(planes * self.expansion)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
         self.stride = stride
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x
-        out = self.relu(self.bn1(self.conv1(x)))
-        out = self.relu(self.bn2(self.conv2(out)))
-        out = self.bn3(self.conv3(out))
+        out = bn_act(self.bn1, self.conv1(x), relu=True)
+        out = bn_act(self.bn2, self.conv2(out), relu=True)
+        out = self.conv3(out)
         if self.downsample is not None:
             identity = self.downsample(x)
-        out += identity
-        return self.relu(out)
+        return bn_act(self.bn3, out, residual=identity, relu=True)
 
 
 class ResNet(nn.Module):
@@ -107,7 +107,7 @@ class ResNet(nn.Module):
                                   kernel_size=7, stride=2, padding=3,
                                   bias=False)
             self.maxpool = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
-        self.bn1 = nn.BatchNorm2d(64)
+        self.bn1 = FusedBatchNorm2d(64)
         self.relu = nn.ReLU(inplace=True)
         self.layer1 = self._make_layer(block, 64, layers[0])
         self.layer2 = self._make_layer(block, 128, layers[1], stride=2)
@@ -130,7 +130,7 @@ class ResNet(nn.Module):
         if stride != 1 or self.inplanes != planes * block.expansion:
             downsample = nn.Sequential(
                 conv1x1(self.inplanes, planes * block.expansion, stride),
-                nn.BatchNorm2d(planes * block.expansion),
+                FusedBatchNorm2d(planes * block.expansion),
             )
         layers = [block(self.inplanes, planes, stride, downsample)]
         self.inplanes = planes * block.expansion
@@ -138,7 +138,7 @@ class ResNet(nn.Module):
         return nn.Sequential(*layers)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.maxpool(bn_act(self.bn1, self.conv1(x), relu=True))
         x = self.layer1(x)
         x = self.layer2(x)
         x = self.layer3(x)

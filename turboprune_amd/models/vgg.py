@@ -13,6 +13,7 @@ from typing import List, Union
 import torch
 import torch.nn as nn
 
+from turboprune_amd.ops.bn import FusedBatchNorm2d
 from turboprune_amd.ops.mask_layers import Conv1dMask, ConvMask
 
 _CFGS = {
@@ -34,7 +35,7 @@ def _make_features(cfg: List[Union[int, str]], batch_norm: bool) -> nn.Sequentia
             layers.append(ConvMask(in_channels=in_ch, out_channels=int(v),
                                    kernel_size=3, padding=1, bias=True))
             if batch_norm:
-                layers.append(nn.BatchNorm2d(int(v)))
+                layers.append(FusedBatchNorm2d(int(v)))
             layers.append(nn.ReLU(inplace=True))
             in_ch = int(v)
     return nn.Sequential(*layers)
