@@ -13,9 +13,17 @@
 //                [finalize] dgamma/dbeta + dx coefficients  ->
 //                [apply] dx (+ dres) — ONE elementwise pass
 //
-// Layout: NHWC (channels_last) — adjacent lanes read adjacent channels,
-// fully coalesced; rows (N*H*W) are grid-strided. Stats/params fp32,
-// activations bf16 or fp32.
+// Layout: NHWC (channels_last): the channel axis is fastest, so every
+// access is a 16-byte (8 x bf16 / 4 x f32 "octet") vector load — scalar
+// bf16 loads are 2-2.5x slower on CDNA4 (guide §6 G13). Reductions use
+// a (row-blocks x channel-blocks) grid, 8 channel-octets x 32 row-lanes
+// per 256-thread block, LDS tree over row-lanes, one atomicAdd per
+// channel per block. Elementwise passes are flat 16B grid-stride with
+// per-channel coefficients cached in LDS. Stats/params fp32.
+//
+// Fast path requires C % 8 == 0 (bf16) / C % 4 == 0 (f32) — true for
+// every model in the zoo; other C fall back to torch composed ops at the
+// Python layer.
 
 #include <ATen/ATen.h>
 #include <ATen/hip/HIPContext.h>
@@ -24,21 +32,97 @@
 
 namespace turboprune {
 
-// ---------------- forward reduce: per-channel sum & sumsq ----------------
+// element vector: 16 bytes of activations
+template <typename T>
+struct Octet;
+template <>
+struct Octet<__hip_bfloat16> {
+  static constexpr int kN = 8;
+  uint4 raw;
+  TP_DEVICE float get(int j) const {
+    const __hip_bfloat16* p = reinterpret_cast<const __hip_bfloat16*>(&raw);
+    return __bfloat162float(p[j]);
+  }
+  TP_DEVICE void set(int j, float v) {
+    __hip_bfloat16* p = reinterpret_cast<__hip_bfloat16*>(&raw);
+    p[j] = __float2bfloat16(v);
+  }
+};
+template <>
+struct Octet<float> {
+  static constexpr int kN = 4;
+  uint4 raw;
+  TP_DEVICE float get(int j) const {
+    return reinterpret_cast<const float*>(&raw)[j];
+  }
+  TP_DEVICE void set(int j, float v) {
+    reinterpret_cast<float*>(&raw)[j] = v;
+  }
+};
+
+template <typename T>
+TP_DEVICE Octet<T> load_octet(const T* p) {
+  Octet<T> o;
+  o.raw = *reinterpret_cast<const uint4*>(p);
+  return o;
+}
+template <typename T>
+TP_DEVICE void store_octet(T* p, const Octet<T>& o) {
+  *reinterpret_cast<uint4*>(p) = o.raw;
+}
+
+// ---------------- forward reduce ----------------------------------------
+// grid: (row_blocks, channel_blocks); block: 256 = 8 octets x 32 lanes
 template <typename T>
 __global__ void bn_reduce_kernel(const T* __restrict__ x, int64_t rows,
                                  int C, float* __restrict__ sum,
                                  float* __restrict__ sumsq) {
-  int c = blockIdx.y * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float s = 0.f, ss = 0.f;
-  for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
-    float v = to_float<T>(x[r * C + c]);
-    s += v;
-    ss += v * v;
+  constexpr int VN = Octet<T>::kN;
+  __shared__ float lsum[8][Octet<T>::kN][32];
+  __shared__ float lss[8][Octet<T>::kN][32];
+  int oct_in_blk = threadIdx.x & 7;   // 8 octets per block
+  int lane = threadIdx.x >> 3;        // 32 row-lanes
+  int oct = blockIdx.y * 8 + oct_in_blk;
+  int c0 = oct * VN;
+  float s[VN], ss[VN];
+#pragma unroll
+  for (int j = 0; j < VN; ++j) s[j] = ss[j] = 0.f;
+  if (c0 < C) {
+    for (int64_t r = (int64_t)blockIdx.x * 32 + lane; r < rows;
+         r += (int64_t)gridDim.x * 32) {
+      Octet<T> o = load_octet(x + r * C + c0);
+#pragma unroll
+      for (int j = 0; j < VN; ++j) {
+        float v = o.get(j);
+        s[j] += v;
+        ss[j] += v * v;
+      }
+    }
   }
-  atomicAdd(&sum[c], s);
-  atomicAdd(&sumsq[c], ss);
+#pragma unroll
+  for (int j = 0; j < VN; ++j) {
+    lsum[oct_in_blk][j][lane] = s[j];
+    lss[oct_in_blk][j][lane] = ss[j];
+  }
+  __syncthreads();
+  // tree over the 32 lanes: threads with lane<16 fold
+  for (int step = 16; step > 0; step >>= 1) {
+    if (lane < step) {
+#pragma unroll
+      for (int j = 0; j < VN; ++j) {
+        lsum[oct_in_blk][j][lane] += lsum[oct_in_blk][j][lane + step];
+        lss[oct_in_blk][j][lane] += lss[oct_in_blk][j][lane + step];
+      }
+    }
+    __syncthreads();
+  }
+  if (lane == 0 && c0 < C) {
+#pragma unroll
+    for (int j = 0; j < VN; ++j) {
+      atomicAdd(&sum[c0 + j], lsum[oct_in_blk][j][0]);
+      atomicAdd(&sumsq[c0 + j], lss[oct_in_blk][j][0]);
+    }
+  }
 }
 
 // ---------------- forward finalize (one small launch) --------------------
@@ -86,7 +170,8 @@ __global__ void bn_eval_coeffs_kernel(const float* __restrict__ running_mean,
   shift[c] = beta[c] - running_mean[c] * sc;
 }
 
-// ---------------- forward apply: y = relu?(x*scale+shift (+res)) ---------
+// ---------------- forward apply ------------------------------------------
+// flat grid-stride over 16B octets; scale/shift cached in LDS (C <= 4096)
 template <typename T, bool RELU, bool RES>
 __global__ void bn_apply_kernel(const T* __restrict__ x,
                                 const T* __restrict__ res,
@@ -94,15 +179,34 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
                                 const float* __restrict__ scale,
                                 const float* __restrict__ shift,
                                 int64_t rows, int C) {
-  int64_t total = rows * C;
+  constexpr int VN = Octet<T>::kN;
+  extern __shared__ __attribute__((aligned(16))) float lds_coeff[];
+  float* lscale = lds_coeff;
+  float* lshift = lds_coeff + C;
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    lscale[c] = scale[c];
+    lshift[c] = shift[c];
+  }
+  __syncthreads();
+  int n_oct = C / VN;
+  int64_t total = rows * n_oct;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += stride) {
-    int c = (int)(i % C);
-    float v = to_float<T>(x[i]) * scale[c] + shift[c];
-    if (RES) v += to_float<T>(res[i]);
-    if (RELU) v = fmaxf(v, 0.f);
-    y[i] = from_float<T>(v);
+    int c0 = (int)(i % n_oct) * VN;
+    int64_t base = (i / n_oct) * C + c0;
+    Octet<T> o = load_octet(x + base);
+    Octet<T> r;
+    if (RES) r = load_octet(res + base);
+    Octet<T> out;
+#pragma unroll
+    for (int j = 0; j < VN; ++j) {
+      float v = o.get(j) * lscale[c0 + j] + lshift[c0 + j];
+      if (RES) v += r.get(j);
+      if (RELU) v = fmaxf(v, 0.f);
+      out.set(j, v);
+    }
+    store_octet(y + base, out);
   }
 }
 
@@ -116,19 +220,61 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
                                      const float* __restrict__ save_rstd,
                                      float* __restrict__ sum_dy,
                                      float* __restrict__ sum_dy_xhat) {
-  int c = blockIdx.y * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float mean = save_mean[c], rstd = save_rstd[c];
-  float s = 0.f, sx = 0.f;
-  for (int64_t r = blockIdx.x; r < rows; r += gridDim.x) {
-    int64_t i = r * C + c;
-    float g = to_float<T>(dy[i]);
-    if (RELU && to_float<T>(y[i]) <= 0.f) g = 0.f;
-    s += g;
-    sx += g * (to_float<T>(x[i]) - mean) * rstd;
+  constexpr int VN = Octet<T>::kN;
+  __shared__ float lsum[8][Octet<T>::kN][32];
+  __shared__ float lsx[8][Octet<T>::kN][32];
+  int oct_in_blk = threadIdx.x & 7;
+  int lane = threadIdx.x >> 3;
+  int oct = blockIdx.y * 8 + oct_in_blk;
+  int c0 = oct * VN;
+  float s[VN], sx[VN], mean[VN], rstd[VN];
+#pragma unroll
+  for (int j = 0; j < VN; ++j) s[j] = sx[j] = 0.f;
+  if (c0 < C) {
+#pragma unroll
+    for (int j = 0; j < VN; ++j) {
+      mean[j] = save_mean[c0 + j];
+      rstd[j] = save_rstd[c0 + j];
+    }
+    for (int64_t r = (int64_t)blockIdx.x * 32 + lane; r < rows;
+         r += (int64_t)gridDim.x * 32) {
+      int64_t base = r * C + c0;
+      Octet<T> ox = load_octet(x + base);
+      Octet<T> og = load_octet(dy + base);
+      Octet<T> oy;
+      if (RELU) oy = load_octet(y + base);
+#pragma unroll
+      for (int j = 0; j < VN; ++j) {
+        float g = og.get(j);
+        if (RELU && oy.get(j) <= 0.f) g = 0.f;
+        s[j] += g;
+        sx[j] += g * (ox.get(j) - mean[j]) * rstd[j];
+      }
+    }
   }
-  atomicAdd(&sum_dy[c], s);
-  atomicAdd(&sum_dy_xhat[c], sx);
+#pragma unroll
+  for (int j = 0; j < VN; ++j) {
+    lsum[oct_in_blk][j][lane] = s[j];
+    lsx[oct_in_blk][j][lane] = sx[j];
+  }
+  __syncthreads();
+  for (int step = 16; step > 0; step >>= 1) {
+    if (lane < step) {
+#pragma unroll
+      for (int j = 0; j < VN; ++j) {
+        lsum[oct_in_blk][j][lane] += lsum[oct_in_blk][j][lane + step];
+        lsx[oct_in_blk][j][lane] += lsx[oct_in_blk][j][lane + step];
+      }
+    }
+    __syncthreads();
+  }
+  if (lane == 0 && c0 < C) {
+#pragma unroll
+    for (int j = 0; j < VN; ++j) {
+      atomicAdd(&sum_dy[c0 + j], lsum[oct_in_blk][j][0]);
+      atomicAdd(&sum_dy_xhat[c0 + j], lsx[oct_in_blk][j][0]);
+    }
+  }
 }
 
 __global__ void bn_bwd_finalize_kernel(const float* __restrict__ sum_dy,
@@ -152,6 +298,8 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ sum_dy,
   c_const[c] = -gr * inv_n * sum_dy[c];
 }
 
+// ---------------- backward apply -----------------------------------------
+// LDS caches 5 coeff arrays: mean, rstd, c_dy, c_xhat, c_const
 template <typename T, bool RELU, bool RES>
 __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
                                     const T* __restrict__ y,
@@ -164,16 +312,44 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
                                     const float* __restrict__ c_xhat,
                                     const float* __restrict__ c_const,
                                     int64_t rows, int C) {
-  int64_t total = rows * C;
+  constexpr int VN = Octet<T>::kN;
+  extern __shared__ __attribute__((aligned(16))) float lds_coeff[];
+  float* lmean = lds_coeff;
+  float* lrstd = lds_coeff + C;
+  float* lcdy = lds_coeff + 2 * C;
+  float* lcx = lds_coeff + 3 * C;
+  float* lcc = lds_coeff + 4 * C;
+  for (int c = threadIdx.x; c < C; c += blockDim.x) {
+    lmean[c] = save_mean[c];
+    lrstd[c] = save_rstd[c];
+    lcdy[c] = c_dy[c];
+    lcx[c] = c_xhat[c];
+    lcc[c] = c_const[c];
+  }
+  __syncthreads();
+  int n_oct = C / VN;
+  int64_t total = rows * n_oct;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
   for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
        i += stride) {
-    int c = (int)(i % C);
-    float g = to_float<T>(dy[i]);
-    if (RELU && to_float<T>(y[i]) <= 0.f) g = 0.f;
-    if (RES) dres[i] = from_float<T>(g);
-    float xhat = (to_float<T>(x[i]) - save_mean[c]) * save_rstd[c];
-    dx[i] = from_float<T>(c_dy[c] * g + c_xhat[c] * xhat + c_const[c]);
+    int c0 = (int)(i % n_oct) * VN;
+    int64_t base = (i / n_oct) * C + c0;
+    Octet<T> ox = load_octet(x + base);
+    Octet<T> og = load_octet(dy + base);
+    Octet<T> oy;
+    if (RELU) oy = load_octet(y + base);
+    Octet<T> odx, odr;
+#pragma unroll
+    for (int j = 0; j < VN; ++j) {
+      int c = c0 + j;
+      float g = og.get(j);
+      if (RELU && oy.get(j) <= 0.f) g = 0.f;
+      if (RES) odr.set(j, g);
+      float xhat = (ox.get(j) - lmean[c]) * lrstd[c];
+      odx.set(j, lcdy[c] * g + lcx[c] * xhat + lcc[c]);
+    }
+    store_octet(dx + base, odx);
+    if (RES) store_octet(dres + base, odr);
   }
 }
 
@@ -190,10 +366,21 @@ static BNShape bn_shape(const at::Tensor& x) {
   return {x.size(0) * x.size(2) * x.size(3), (int)x.size(1)};
 }
 
+template <typename T>
 static dim3 reduce_grid(int64_t rows, int C) {
-  int cy = (C + kBlock - 1) / kBlock;
-  int rx = (int)std::min<int64_t>((rows + 63) / 64, 1024);
-  return dim3(rx, cy);
+  int cb = (C / Octet<T>::kN + 7) / 8;   // channel blocks (8 octets each)
+  int rb = (int)std::min<int64_t>((rows + 31) / 32, 768);
+  return dim3(rb, cb);
+}
+
+static int apply_grid(int64_t rows, int C, int vn) {
+  return elementwise_grid(rows * (C / vn), kBlock, 4);
+}
+
+bool bn_fast_path_ok(const at::Tensor& x) {
+  int C = (int)x.size(1);
+  int vn = x.scalar_type() == at::kBFloat16 ? 8 : 4;
+  return C % vn == 0 && C <= 4096;
 }
 
 // returns (y, save_mean, save_rstd)
@@ -203,6 +390,8 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd(
     at::Tensor running_mean, at::Tensor running_var, bool training,
     double momentum, double eps, bool relu) {
   auto sh = bn_shape(x);
+  TORCH_CHECK(bn_fast_path_ok(x), "bn_fwd: C must be a multiple of the "
+              "16B vector and <= 4096");
   auto fopt = x.options().dtype(at::kFloat);
   auto scale = at::empty({sh.C}, fopt);
   auto shift = at::empty({sh.C}, fopt);
@@ -216,20 +405,23 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd(
 
 #define BN_DISPATCH_T(fn, ...)                                           \
   if (x.scalar_type() == at::kBFloat16) {                                \
-    fn(__hip_bfloat16, __VA_ARGS__);                                     \
+    using T = __hip_bfloat16;                                            \
+    fn(__VA_ARGS__);                                                     \
   } else {                                                               \
-    fn(float, __VA_ARGS__);                                              \
+    using T = float;                                                     \
+    fn(__VA_ARGS__);                                                     \
   }
 
   if (training) {
     auto sum = at::zeros({sh.C}, fopt);
     auto sumsq = at::zeros({sh.C}, fopt);
-#define BN_RED(T, ...)                                                    \
-    hipLaunchKernelGGL((bn_reduce_kernel<T>), reduce_grid(sh.rows, sh.C), \
-                       dim3(kBlock), 0, stream,                           \
-                       reinterpret_cast<const T*>(x.data_ptr()), sh.rows, \
-                       sh.C, sum.data_ptr<float>(), sumsq.data_ptr<float>())
-    BN_DISPATCH_T(BN_RED, )
+#define BN_RED()                                                          \
+    hipLaunchKernelGGL((bn_reduce_kernel<T>),                             \
+                       (reduce_grid<T>(sh.rows, sh.C)), dim3(kBlock), 0,  \
+                       stream, reinterpret_cast<const T*>(x.data_ptr()),  \
+                       sh.rows, sh.C, sum.data_ptr<float>(),              \
+                       sumsq.data_ptr<float>())
+    BN_DISPATCH_T(BN_RED)
 #undef BN_RED
     float inv_n = 1.0f / (float)sh.rows;
     float unbiased = sh.rows > 1 ? (float)sh.rows / (sh.rows - 1) : 1.0f;
@@ -259,23 +451,21 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd(
     res = residual->contiguous(at::MemoryFormat::ChannelsLast);
     TORCH_CHECK(res.sizes() == x.sizes());
   }
-  int64_t total = sh.rows * sh.C;
-  int grid = elementwise_grid(total, kBlock, 8);
-#define BN_APPLY(T, RELUF, RESF)                                          \
-  hipLaunchKernelGGL((bn_apply_kernel<T, RELUF, RESF>), dim3(grid),       \
-                     dim3(kBlock), 0, stream,                             \
-                     reinterpret_cast<const T*>(x.data_ptr()),            \
-                     RESF ? reinterpret_cast<const T*>(res.data_ptr())    \
-                          : nullptr,                                      \
-                     reinterpret_cast<T*>(y.data_ptr()),                  \
-                     scale.data_ptr<float>(), shift.data_ptr<float>(),    \
-                     sh.rows, sh.C)
-#define BN_APPLY_D(T, unused)                                             \
-  if (relu) { if (has_res) BN_APPLY(T, true, true);                       \
-              else BN_APPLY(T, true, false); }                            \
-  else { if (has_res) BN_APPLY(T, false, true);                           \
-         else BN_APPLY(T, false, false); }
-  BN_DISPATCH_T(BN_APPLY_D, )
+  size_t lds_bytes = 2 * sh.C * sizeof(float);
+#define BN_APPLY(RELUF, RESF)                                             \
+  hipLaunchKernelGGL(                                                     \
+      (bn_apply_kernel<T, RELUF, RESF>),                                  \
+      dim3(apply_grid(sh.rows, sh.C, Octet<T>::kN)), dim3(kBlock),        \
+      lds_bytes, stream, reinterpret_cast<const T*>(x.data_ptr()),        \
+      RESF ? reinterpret_cast<const T*>(res.data_ptr()) : nullptr,        \
+      reinterpret_cast<T*>(y.data_ptr()), scale.data_ptr<float>(),        \
+      shift.data_ptr<float>(), sh.rows, sh.C)
+#define BN_APPLY_D()                                                      \
+  if (relu) { if (has_res) BN_APPLY(true, true);                          \
+              else BN_APPLY(true, false); }                               \
+  else { if (has_res) BN_APPLY(false, true);                              \
+         else BN_APPLY(false, false); }
+  BN_DISPATCH_T(BN_APPLY_D)
 #undef BN_APPLY_D
 #undef BN_APPLY
   return {y, save_mean, save_rstd};
@@ -294,19 +484,19 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd(
 
   auto sum_dy = at::zeros({sh.C}, fopt);
   auto sum_dy_xhat = at::zeros({sh.C}, fopt);
-#define BN_BRED(T, RELUF)                                                  \
+#define BN_BRED(RELUF)                                                     \
   hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, RELUF>),                     \
-                     reduce_grid(sh.rows, sh.C), dim3(kBlock), 0, stream,  \
-                     reinterpret_cast<const T*>(x.data_ptr()),             \
+                     (reduce_grid<T>(sh.rows, sh.C)), dim3(kBlock), 0,     \
+                     stream, reinterpret_cast<const T*>(x.data_ptr()),     \
                      reinterpret_cast<const T*>(y.data_ptr()),             \
                      reinterpret_cast<const T*>(dy.data_ptr()), sh.rows,   \
                      sh.C, save_mean.data_ptr<float>(),                    \
                      save_rstd.data_ptr<float>(),                          \
                      sum_dy.data_ptr<float>(),                            \
                      sum_dy_xhat.data_ptr<float>())
-#define BN_BRED_D(T, unused)                                               \
-  if (relu) BN_BRED(T, true); else BN_BRED(T, false);
-  BN_DISPATCH_T(BN_BRED_D, )
+#define BN_BRED_D()                                                        \
+  if (relu) BN_BRED(true); else BN_BRED(false);
+  BN_DISPATCH_T(BN_BRED_D)
 #undef BN_BRED_D
 #undef BN_BRED
 
@@ -328,27 +518,25 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd(
   auto dx = at::empty_like(x);
   at::Tensor dres;
   if (has_res) dres = at::empty_like(x);
-  int64_t total = sh.rows * sh.C;
-  int grid = elementwise_grid(total, kBlock, 8);
-#define BN_BAPPLY(T, RELUF, RESF)                                          \
-  hipLaunchKernelGGL((bn_bwd_apply_kernel<T, RELUF, RESF>), dim3(grid),    \
-                     dim3(kBlock), 0, stream,                              \
-                     reinterpret_cast<const T*>(x.data_ptr()),             \
-                     reinterpret_cast<const T*>(y.data_ptr()),             \
-                     reinterpret_cast<const T*>(dy.data_ptr()),            \
-                     reinterpret_cast<T*>(dx.data_ptr()),                  \
-                     RESF ? reinterpret_cast<T*>(dres.data_ptr())          \
-                          : nullptr,                                       \
-                     save_mean.data_ptr<float>(),                          \
-                     save_rstd.data_ptr<float>(), c_dy.data_ptr<float>(),  \
-                     c_xhat.data_ptr<float>(), c_const.data_ptr<float>(),  \
-                     sh.rows, sh.C)
-#define BN_BAPPLY_D(T, unused)                                             \
-  if (relu) { if (has_res) BN_BAPPLY(T, true, true);                       \
-              else BN_BAPPLY(T, true, false); }                            \
-  else { if (has_res) BN_BAPPLY(T, false, true);                           \
-         else BN_BAPPLY(T, false, false); }
-  BN_DISPATCH_T(BN_BAPPLY_D, )
+  size_t lds_bytes = 5 * sh.C * sizeof(float);
+#define BN_BAPPLY(RELUF, RESF)                                             \
+  hipLaunchKernelGGL(                                                      \
+      (bn_bwd_apply_kernel<T, RELUF, RESF>),                               \
+      dim3(apply_grid(sh.rows, sh.C, Octet<T>::kN)), dim3(kBlock),         \
+      lds_bytes, stream, reinterpret_cast<const T*>(x.data_ptr()),         \
+      reinterpret_cast<const T*>(y.data_ptr()),                            \
+      reinterpret_cast<const T*>(dy.data_ptr()),                           \
+      reinterpret_cast<T*>(dx.data_ptr()),                                 \
+      RESF ? reinterpret_cast<T*>(dres.data_ptr()) : nullptr,              \
+      save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),            \
+      c_dy.data_ptr<float>(), c_xhat.data_ptr<float>(),                    \
+      c_const.data_ptr<float>(), sh.rows, sh.C)
+#define BN_BAPPLY_D()                                                      \
+  if (relu) { if (has_res) BN_BAPPLY(true, true);                          \
+              else BN_BAPPLY(true, false); }                               \
+  else { if (has_res) BN_BAPPLY(false, true);                              \
+         else BN_BAPPLY(false, false); }
+  BN_DISPATCH_T(BN_BAPPLY_D)
 #undef BN_BAPPLY_D
 #undef BN_BAPPLY
   return {dx, dgamma, dbeta, dres};

@@ -75,6 +75,8 @@ def bn_act(bn: nn.BatchNorm2d, x: torch.Tensor,
         and (bn.training or not torch.is_grad_enabled())
         and _backend.use_native(x)
     )
+    if use_fused:
+        use_fused = _backend.extension().bn_fast_path_ok(x)
     if bn.training and bn.track_running_stats \
             and bn.num_batches_tracked is not None:
         bn.num_batches_tracked.add_(1)
