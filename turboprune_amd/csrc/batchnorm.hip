@@ -88,8 +88,24 @@ __global__ void bn_reduce_kernel(const T* __restrict__ x, int64_t rows,
 #pragma unroll
   for (int j = 0; j < VN; ++j) s[j] = ss[j] = 0.f;
   if (c0 < C) {
-    for (int64_t r = (int64_t)blockIdx.x * 32 + lane; r < rows;
-         r += (int64_t)gridDim.x * 32) {
+    int64_t r = (int64_t)blockIdx.x * 32 + lane;
+    int64_t rstep = (int64_t)gridDim.x * 32;
+    // 4-deep unroll: keep 4 independent 16B loads in flight
+    // (single-load loop was HBM-latency-bound at ~2.4 TB/s)
+    for (; r + 3 * rstep < rows; r += 4 * rstep) {
+      Octet<T> o0 = load_octet(x + r * C + c0);
+      Octet<T> o1 = load_octet(x + (r + rstep) * C + c0);
+      Octet<T> o2 = load_octet(x + (r + 2 * rstep) * C + c0);
+      Octet<T> o3 = load_octet(x + (r + 3 * rstep) * C + c0);
+#pragma unroll
+      for (int j = 0; j < VN; ++j) {
+        float v0 = o0.get(j), v1 = o1.get(j), v2 = o2.get(j),
+              v3 = o3.get(j);
+        s[j] += (v0 + v1) + (v2 + v3);
+        ss[j] += (v0 * v0 + v1 * v1) + (v2 * v2 + v3 * v3);
+      }
+    }
+    for (; r < rows; r += rstep) {
       Octet<T> o = load_octet(x + r * C + c0);
 #pragma unroll
       for (int j = 0; j < VN; ++j) {
@@ -236,8 +252,25 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
       mean[j] = save_mean[c0 + j];
       rstd[j] = save_rstd[c0 + j];
     }
-    for (int64_t r = (int64_t)blockIdx.x * 32 + lane; r < rows;
-         r += (int64_t)gridDim.x * 32) {
+    int64_t r = (int64_t)blockIdx.x * 32 + lane;
+    int64_t rstep = (int64_t)gridDim.x * 32;
+    for (; r + rstep < rows; r += 2 * rstep) {
+      int64_t b0 = r * C + c0, b1 = (r + rstep) * C + c0;
+      Octet<T> ox0 = load_octet(x + b0), ox1 = load_octet(x + b1);
+      Octet<T> og0 = load_octet(dy + b0), og1 = load_octet(dy + b1);
+      Octet<T> oy0, oy1;
+      if (RELU) { oy0 = load_octet(y + b0); oy1 = load_octet(y + b1); }
+#pragma unroll
+      for (int j = 0; j < VN; ++j) {
+        float g0 = og0.get(j), g1 = og1.get(j);
+        if (RELU && oy0.get(j) <= 0.f) g0 = 0.f;
+        if (RELU && oy1.get(j) <= 0.f) g1 = 0.f;
+        s[j] += g0 + g1;
+        sx[j] += g0 * (ox0.get(j) - mean[j]) * rstd[j] +
+                 g1 * (ox1.get(j) - mean[j]) * rstd[j];
+      }
+    }
+    for (; r < rows; r += rstep) {
       int64_t base = r * C + c0;
       Octet<T> ox = load_octet(x + base);
       Octet<T> og = load_octet(dy + base);
@@ -369,12 +402,19 @@ static BNShape bn_shape(const at::Tensor& x) {
 template <typename T>
 static dim3 reduce_grid(int64_t rows, int C) {
   int cb = (C / Octet<T>::kN + 7) / 8;   // channel blocks (8 octets each)
-  int rb = (int)std::min<int64_t>((rows + 31) / 32, 768);
+  // ~16 rows per thread; cap total blocks near 4k for >>256-CU fill
+  int64_t rb_want = (rows + 32 * 16 - 1) / (32 * 16);
+  int rb = (int)std::min<int64_t>(std::max<int64_t>(rb_want, 1),
+                                  std::max<int64_t>(4096 / cb, 8));
   return dim3(rb, cb);
 }
 
 static int apply_grid(int64_t rows, int C, int vn) {
-  return elementwise_grid(rows * (C / vn), kBlock, 4);
+  // ~2 octets per thread, up to 8192 blocks: full-chip TLP to hide HBM
+  // latency (2048-block cap left these passes latency-bound)
+  int64_t total = rows * (C / vn);
+  int64_t blocks = (total + kBlock * 2 - 1) / (kBlock * 2);
+  return (int)std::min<int64_t>(std::max<int64_t>(blocks, 1), 8192);
 }
 
 bool bn_fast_path_ok(const at::Tensor& x) {
