@@ -3,11 +3,14 @@
 shapes. Prints per-shape times and effective GB/s."""
 
 import json
+import os
+import sys
 import time
 
 import torch
 
-from turboprune_amd.ops._backend import extension
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+from turboprune_amd.ops._backend import extension  # noqa: E402
 
 SHAPES = [(64, 112), (64, 56), (256, 56), (128, 56), (128, 28), (512, 28),
           (256, 28), (256, 14), (1024, 14), (512, 14), (512, 7), (2048, 7)]
