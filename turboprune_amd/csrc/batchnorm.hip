@@ -72,48 +72,91 @@ TP_DEVICE void store_octet(T* p, const Octet<T>& o) {
 }
 
 // ---------------- forward reduce ----------------------------------------
-// grid: (row_blocks, channel_blocks); block: 256 = 8 octets x 32 lanes
+// grid: (row_blocks, channel_blocks); block: 256 = 8 octets x 32 lanes.
+//
+// VALU budget: at 6.3 TB/s each CU must retire a 16B octet every ~6 VALU
+// issue slots, so the inner loop uses explicit bit ops (bf16 -> f32 is a
+// shift / a mask, no v_cvt) and packed f32x2 accumulators (v_pk_add_f32
+// / v_pk_fma_f32): 2 shifts + 1 pk_add + 1 pk_fma per dword (2 elems).
+typedef __attribute__((ext_vector_type(2))) float f32x2;
+
+TP_DEVICE f32x2 bf16pair_to_f32x2(uint32_t d) {
+  f32x2 v;
+  v[0] = __uint_as_float(d << 16);
+  v[1] = __uint_as_float(d & 0xffff0000u);
+  return v;
+}
+
+// per-dword accumulate helpers (sum += v, acc2 += v*w)
+template <typename T>
+struct OctetAcc;
+template <>
+struct OctetAcc<__hip_bfloat16> {
+  static constexpr int kP = 4;  // f32x2 pairs per octet
+  TP_DEVICE static void unpack(const Octet<__hip_bfloat16>& o,
+                               f32x2 (&v)[4]) {
+    const uint32_t* d = reinterpret_cast<const uint32_t*>(&o.raw);
+#pragma unroll
+    for (int p = 0; p < 4; ++p) v[p] = bf16pair_to_f32x2(d[p]);
+  }
+  // channel index of pair p, slot s: 2*p + s
+};
+template <>
+struct OctetAcc<float> {
+  static constexpr int kP = 2;
+  TP_DEVICE static void unpack(const Octet<float>& o, f32x2 (&v)[2]) {
+    const float* f = reinterpret_cast<const float*>(&o.raw);
+    v[0][0] = f[0]; v[0][1] = f[1];
+    v[1][0] = f[2]; v[1][1] = f[3];
+  }
+};
+
 template <typename T>
 __global__ void bn_reduce_kernel(const T* __restrict__ x, int64_t rows,
                                  int C, float* __restrict__ sum,
                                  float* __restrict__ sumsq) {
   constexpr int VN = Octet<T>::kN;
+  constexpr int P = OctetAcc<T>::kP;
   __shared__ float lsum[8][Octet<T>::kN][32];
   __shared__ float lss[8][Octet<T>::kN][32];
   int oct_in_blk = threadIdx.x & 7;   // 8 octets per block
   int lane = threadIdx.x >> 3;        // 32 row-lanes
   int oct = blockIdx.y * 8 + oct_in_blk;
   int c0 = oct * VN;
-  float s[VN], ss[VN];
+  f32x2 s2[P], ss2[P];
 #pragma unroll
-  for (int j = 0; j < VN; ++j) s[j] = ss[j] = 0.f;
+  for (int p = 0; p < P; ++p) s2[p] = ss2[p] = f32x2{0.f, 0.f};
   if (c0 < C) {
     int64_t r = (int64_t)blockIdx.x * 32 + lane;
     int64_t rstep = (int64_t)gridDim.x * 32;
-    // 4-deep unroll: keep 4 independent 16B loads in flight
-    // (single-load loop was HBM-latency-bound at ~2.4 TB/s)
-    for (; r + 3 * rstep < rows; r += 4 * rstep) {
+    for (; r + rstep < rows; r += 2 * rstep) {
       Octet<T> o0 = load_octet(x + r * C + c0);
       Octet<T> o1 = load_octet(x + (r + rstep) * C + c0);
-      Octet<T> o2 = load_octet(x + (r + 2 * rstep) * C + c0);
-      Octet<T> o3 = load_octet(x + (r + 3 * rstep) * C + c0);
+      f32x2 v0[P], v1[P];
+      OctetAcc<T>::unpack(o0, v0);
+      OctetAcc<T>::unpack(o1, v1);
 #pragma unroll
-      for (int j = 0; j < VN; ++j) {
-        float v0 = o0.get(j), v1 = o1.get(j), v2 = o2.get(j),
-              v3 = o3.get(j);
-        s[j] += (v0 + v1) + (v2 + v3);
-        ss[j] += (v0 * v0 + v1 * v1) + (v2 * v2 + v3 * v3);
+      for (int p = 0; p < P; ++p) {
+        s2[p] += v0[p] + v1[p];
+        ss2[p] += v0[p] * v0[p] + v1[p] * v1[p];
       }
     }
     for (; r < rows; r += rstep) {
       Octet<T> o = load_octet(x + r * C + c0);
+      f32x2 v[P];
+      OctetAcc<T>::unpack(o, v);
 #pragma unroll
-      for (int j = 0; j < VN; ++j) {
-        float v = o.get(j);
-        s[j] += v;
-        ss[j] += v * v;
+      for (int p = 0; p < P; ++p) {
+        s2[p] += v[p];
+        ss2[p] += v[p] * v[p];
       }
     }
+  }
+  float s[VN], ss[VN];
+#pragma unroll
+  for (int p = 0; p < P; ++p) {
+    s[2 * p] = s2[p][0]; s[2 * p + 1] = s2[p][1];
+    ss[2 * p] = ss2[p][0]; ss[2 * p + 1] = ss2[p][1];
   }
 #pragma unroll
   for (int j = 0; j < VN; ++j) {
@@ -227,63 +270,57 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
 }
 
 // ---------------- backward reduce ----------------------------------------
+// accumulates sum_g = Σ dy_eff and sum_gx = Σ dy_eff * x per channel
+// (the xhat projection Σ g·xhat = rstd·(sum_gx − mean·sum_g) is applied
+// in the finalize kernel — keeps the hot loop at ~3 packed VALU/dword).
 template <typename T, bool RELU>
 __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
                                      const T* __restrict__ y,
                                      const T* __restrict__ dy, int64_t rows,
                                      int C,
-                                     const float* __restrict__ save_mean,
-                                     const float* __restrict__ save_rstd,
-                                     float* __restrict__ sum_dy,
-                                     float* __restrict__ sum_dy_xhat) {
+                                     float* __restrict__ sum_g,
+                                     float* __restrict__ sum_gx) {
   constexpr int VN = Octet<T>::kN;
+  constexpr int P = OctetAcc<T>::kP;
   __shared__ float lsum[8][Octet<T>::kN][32];
   __shared__ float lsx[8][Octet<T>::kN][32];
   int oct_in_blk = threadIdx.x & 7;
   int lane = threadIdx.x >> 3;
   int oct = blockIdx.y * 8 + oct_in_blk;
   int c0 = oct * VN;
-  float s[VN], sx[VN], mean[VN], rstd[VN];
+  f32x2 s2[P], sx2[P];
 #pragma unroll
-  for (int j = 0; j < VN; ++j) s[j] = sx[j] = 0.f;
+  for (int p = 0; p < P; ++p) s2[p] = sx2[p] = f32x2{0.f, 0.f};
   if (c0 < C) {
-#pragma unroll
-    for (int j = 0; j < VN; ++j) {
-      mean[j] = save_mean[c0 + j];
-      rstd[j] = save_rstd[c0 + j];
-    }
     int64_t r = (int64_t)blockIdx.x * 32 + lane;
     int64_t rstep = (int64_t)gridDim.x * 32;
-    for (; r + rstep < rows; r += 2 * rstep) {
-      int64_t b0 = r * C + c0, b1 = (r + rstep) * C + c0;
-      Octet<T> ox0 = load_octet(x + b0), ox1 = load_octet(x + b1);
-      Octet<T> og0 = load_octet(dy + b0), og1 = load_octet(dy + b1);
-      Octet<T> oy0, oy1;
-      if (RELU) { oy0 = load_octet(y + b0); oy1 = load_octet(y + b1); }
-#pragma unroll
-      for (int j = 0; j < VN; ++j) {
-        float g0 = og0.get(j), g1 = og1.get(j);
-        if (RELU && oy0.get(j) <= 0.f) g0 = 0.f;
-        if (RELU && oy1.get(j) <= 0.f) g1 = 0.f;
-        s[j] += g0 + g1;
-        sx[j] += g0 * (ox0.get(j) - mean[j]) * rstd[j] +
-                 g1 * (ox1.get(j) - mean[j]) * rstd[j];
-      }
-    }
     for (; r < rows; r += rstep) {
       int64_t base = r * C + c0;
       Octet<T> ox = load_octet(x + base);
       Octet<T> og = load_octet(dy + base);
       Octet<T> oy;
       if (RELU) oy = load_octet(y + base);
+      f32x2 vx[P], vg[P], vy[P];
+      OctetAcc<T>::unpack(ox, vx);
+      OctetAcc<T>::unpack(og, vg);
+      if (RELU) OctetAcc<T>::unpack(oy, vy);
 #pragma unroll
-      for (int j = 0; j < VN; ++j) {
-        float g = og.get(j);
-        if (RELU && oy.get(j) <= 0.f) g = 0.f;
-        s[j] += g;
-        sx[j] += g * (ox.get(j) - mean[j]) * rstd[j];
+      for (int p = 0; p < P; ++p) {
+        f32x2 g = vg[p];
+        if (RELU) {
+          g[0] = vy[p][0] > 0.f ? g[0] : 0.f;
+          g[1] = vy[p][1] > 0.f ? g[1] : 0.f;
+        }
+        s2[p] += g;
+        sx2[p] += g * vx[p];
       }
     }
+  }
+  float s[VN], sx[VN];
+#pragma unroll
+  for (int p = 0; p < P; ++p) {
+    s[2 * p] = s2[p][0]; s[2 * p + 1] = s2[p][1];
+    sx[2 * p] = sx2[p][0]; sx[2 * p + 1] = sx2[p][1];
   }
 #pragma unroll
   for (int j = 0; j < VN; ++j) {
@@ -304,15 +341,16 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
   if (lane == 0 && c0 < C) {
 #pragma unroll
     for (int j = 0; j < VN; ++j) {
-      atomicAdd(&sum_dy[c0 + j], lsum[oct_in_blk][j][0]);
-      atomicAdd(&sum_dy_xhat[c0 + j], lsx[oct_in_blk][j][0]);
+      atomicAdd(&sum_g[c0 + j], lsum[oct_in_blk][j][0]);
+      atomicAdd(&sum_gx[c0 + j], lsx[oct_in_blk][j][0]);
     }
   }
 }
 
-__global__ void bn_bwd_finalize_kernel(const float* __restrict__ sum_dy,
-                                       const float* __restrict__ sum_dy_xhat,
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ sum_g,
+                                       const float* __restrict__ sum_gx,
                                        const float* __restrict__ gamma,
+                                       const float* __restrict__ save_mean,
                                        const float* __restrict__ save_rstd,
                                        float* __restrict__ dgamma,
                                        float* __restrict__ dbeta,
@@ -322,13 +360,16 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ sum_dy,
                                        float inv_n) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  dgamma[c] = sum_dy_xhat[c];
-  dbeta[c] = sum_dy[c];
+  float sum_dy = sum_g[c];
+  // Σ g·xhat = rstd · (Σ g·x − mean·Σ g)
+  float sum_dy_xhat = save_rstd[c] * (sum_gx[c] - save_mean[c] * sum_dy);
+  dgamma[c] = sum_dy_xhat;
+  dbeta[c] = sum_dy;
   // dx = g*rstd * (dy_eff - inv_n*sum_dy - xhat*inv_n*sum_dy_xhat)
   float gr = gamma[c] * save_rstd[c];
   c_dy[c] = gr;
-  c_xhat[c] = -gr * inv_n * sum_dy_xhat[c];
-  c_const[c] = -gr * inv_n * sum_dy[c];
+  c_xhat[c] = -gr * inv_n * sum_dy_xhat;
+  c_const[c] = -gr * inv_n * sum_dy;
 }
 
 // ---------------- backward apply -----------------------------------------
@@ -530,9 +571,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd(
                      stream, reinterpret_cast<const T*>(x.data_ptr()),     \
                      reinterpret_cast<const T*>(y.data_ptr()),             \
                      reinterpret_cast<const T*>(dy.data_ptr()), sh.rows,   \
-                     sh.C, save_mean.data_ptr<float>(),                    \
-                     save_rstd.data_ptr<float>(),                          \
-                     sum_dy.data_ptr<float>(),                            \
+                     sh.C, sum_dy.data_ptr<float>(),                       \
                      sum_dy_xhat.data_ptr<float>())
 #define BN_BRED_D()                                                        \
   if (relu) BN_BRED(true); else BN_BRED(false);
@@ -549,7 +588,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd(
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(cblocks), dim3(kBlock), 0,
                      stream, sum_dy.data_ptr<float>(),
                      sum_dy_xhat.data_ptr<float>(), g.data_ptr<float>(),
-                     save_rstd.data_ptr<float>(),
+                     save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                      dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
                      c_dy.data_ptr<float>(), c_xhat.data_ptr<float>(),
                      c_const.data_ptr<float>(), sh.C,
