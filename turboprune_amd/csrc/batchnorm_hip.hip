@@ -112,10 +112,13 @@ struct OctetAcc<float> {
   }
 };
 
+// writes per-block partial sums to partial_sum/partial_ss[blockIdx.x][C]
+// (plain stores; a single global accumulator serializes gridDim.x
+// atomicAdds per channel — a ~400 us tail at 4096 blocks).
 template <typename T>
 __global__ void bn_reduce_kernel(const T* __restrict__ x, int64_t rows,
-                                 int C, float* __restrict__ sum,
-                                 float* __restrict__ sumsq) {
+                                 int C, float* __restrict__ partial_sum,
+                                 float* __restrict__ partial_ss) {
   constexpr int VN = Octet<T>::kN;
   constexpr int P = OctetAcc<T>::kP;
   __shared__ float lsum[8][Octet<T>::kN][32];
@@ -177,17 +180,19 @@ __global__ void bn_reduce_kernel(const T* __restrict__ x, int64_t rows,
     __syncthreads();
   }
   if (lane == 0 && c0 < C) {
+    int64_t row = (int64_t)blockIdx.x * C;
 #pragma unroll
     for (int j = 0; j < VN; ++j) {
-      atomicAdd(&sum[c0 + j], lsum[oct_in_blk][j][0]);
-      atomicAdd(&sumsq[c0 + j], lss[oct_in_blk][j][0]);
+      partial_sum[row + c0 + j] = lsum[oct_in_blk][j][0];
+      partial_ss[row + c0 + j] = lss[oct_in_blk][j][0];
     }
   }
 }
 
 // ---------------- forward finalize (one small launch) --------------------
-__global__ void bn_finalize_kernel(const float* __restrict__ sum,
-                                   const float* __restrict__ sumsq,
+__global__ void bn_finalize_kernel(const float* __restrict__ partial_sum,
+                                   const float* __restrict__ partial_ss,
+                                   int rb,
                                    const float* __restrict__ gamma,
                                    const float* __restrict__ beta,
                                    float* __restrict__ running_mean,
@@ -201,8 +206,27 @@ __global__ void bn_finalize_kernel(const float* __restrict__ sum,
                                    bool update_running) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float mean = sum[c] * inv_n;
-  float var = fmaxf(sumsq[c] * inv_n - mean * mean, 0.f);
+  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
+  float q0 = 0.f, q1 = 0.f, q2 = 0.f, q3 = 0.f;
+  int r = 0;
+  for (; r + 3 < rb; r += 4) {
+    s0 += partial_sum[(int64_t)r * C + c];
+    s1 += partial_sum[(int64_t)(r + 1) * C + c];
+    s2 += partial_sum[(int64_t)(r + 2) * C + c];
+    s3 += partial_sum[(int64_t)(r + 3) * C + c];
+    q0 += partial_ss[(int64_t)r * C + c];
+    q1 += partial_ss[(int64_t)(r + 1) * C + c];
+    q2 += partial_ss[(int64_t)(r + 2) * C + c];
+    q3 += partial_ss[(int64_t)(r + 3) * C + c];
+  }
+  for (; r < rb; ++r) {
+    s0 += partial_sum[(int64_t)r * C + c];
+    q0 += partial_ss[(int64_t)r * C + c];
+  }
+  float sum_c = (s0 + s1) + (s2 + s3);
+  float sumsq_c = (q0 + q1) + (q2 + q3);
+  float mean = sum_c * inv_n;
+  float var = fmaxf(sumsq_c * inv_n - mean * mean, 0.f);
   float rstd = rsqrtf(var + eps);
   save_mean[c] = mean;
   save_rstd[c] = rstd;
@@ -279,8 +303,8 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
                                      const T* __restrict__ y,
                                      const T* __restrict__ dy, int64_t rows,
                                      int C,
-                                     float* __restrict__ sum_g,
-                                     float* __restrict__ sum_gx) {
+                                     float* __restrict__ partial_g,
+                                     float* __restrict__ partial_gx) {
   constexpr int VN = Octet<T>::kN;
   constexpr int P = OctetAcc<T>::kP;
   __shared__ float lsum[8][Octet<T>::kN][32];
@@ -340,16 +364,18 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
     __syncthreads();
   }
   if (lane == 0 && c0 < C) {
+    int64_t row = (int64_t)blockIdx.x * C;
 #pragma unroll
     for (int j = 0; j < VN; ++j) {
-      atomicAdd(&sum_g[c0 + j], lsum[oct_in_blk][j][0]);
-      atomicAdd(&sum_gx[c0 + j], lsx[oct_in_blk][j][0]);
+      partial_g[row + c0 + j] = lsum[oct_in_blk][j][0];
+      partial_gx[row + c0 + j] = lsx[oct_in_blk][j][0];
     }
   }
 }
 
-__global__ void bn_bwd_finalize_kernel(const float* __restrict__ sum_g,
-                                       const float* __restrict__ sum_gx,
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ partial_g,
+                                       const float* __restrict__ partial_gx,
+                                       int rb,
                                        const float* __restrict__ gamma,
                                        const float* __restrict__ save_mean,
                                        const float* __restrict__ save_rstd,
@@ -361,9 +387,28 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ sum_g,
                                        float inv_n) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float sum_dy = sum_g[c];
+  float g0 = 0.f, g1 = 0.f, g2 = 0.f, g3 = 0.f;
+  float x0 = 0.f, x1 = 0.f, x2 = 0.f, x3 = 0.f;
+  int r = 0;
+  for (; r + 3 < rb; r += 4) {
+    g0 += partial_g[(int64_t)r * C + c];
+    g1 += partial_g[(int64_t)(r + 1) * C + c];
+    g2 += partial_g[(int64_t)(r + 2) * C + c];
+    g3 += partial_g[(int64_t)(r + 3) * C + c];
+    x0 += partial_gx[(int64_t)r * C + c];
+    x1 += partial_gx[(int64_t)(r + 1) * C + c];
+    x2 += partial_gx[(int64_t)(r + 2) * C + c];
+    x3 += partial_gx[(int64_t)(r + 3) * C + c];
+  }
+  for (; r < rb; ++r) {
+    g0 += partial_g[(int64_t)r * C + c];
+    x0 += partial_gx[(int64_t)r * C + c];
+  }
+  float sum_g_c = (g0 + g1) + (g2 + g3);
+  float sum_gx_c = (x0 + x1) + (x2 + x3);
+  float sum_dy = sum_g_c;
   // Σ g·xhat = rstd · (Σ g·x − mean·Σ g)
-  float sum_dy_xhat = save_rstd[c] * (sum_gx[c] - save_mean[c] * sum_dy);
+  float sum_dy_xhat = save_rstd[c] * (sum_gx_c - save_mean[c] * sum_dy);
   dgamma[c] = sum_dy_xhat;
   dbeta[c] = sum_dy;
   // dx = g*rstd * (dy_eff - inv_n*sum_dy - xhat*inv_n*sum_dy_xhat)
@@ -495,21 +540,26 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd(
   }
 
   if (training) {
-    auto sum = at::zeros({sh.C}, fopt);
-    auto sumsq = at::zeros({sh.C}, fopt);
+    dim3 rg;
+    if (x.scalar_type() == at::kBFloat16)
+      rg = reduce_grid<__hip_bfloat16>(sh.rows, sh.C);
+    else
+      rg = reduce_grid<float>(sh.rows, sh.C);
+    auto partial_sum = at::empty({(int64_t)rg.x, sh.C}, fopt);
+    auto partial_ss = at::empty({(int64_t)rg.x, sh.C}, fopt);
 #define BN_RED()                                                          \
-    hipLaunchKernelGGL((bn_reduce_kernel<T>),                             \
-                       (reduce_grid<T>(sh.rows, sh.C)), dim3(kBlock), 0,  \
+    hipLaunchKernelGGL((bn_reduce_kernel<T>), rg, dim3(kBlock), 0,        \
                        stream, reinterpret_cast<const T*>(x.data_ptr()),  \
-                       sh.rows, sh.C, sum.data_ptr<float>(),              \
-                       sumsq.data_ptr<float>())
+                       sh.rows, sh.C, partial_sum.data_ptr<float>(),      \
+                       partial_ss.data_ptr<float>())
     BN_DISPATCH_T(BN_RED)
 #undef BN_RED
     float inv_n = 1.0f / (float)sh.rows;
     float unbiased = sh.rows > 1 ? (float)sh.rows / (sh.rows - 1) : 1.0f;
     hipLaunchKernelGGL(bn_finalize_kernel, dim3(cblocks), dim3(kBlock), 0,
-                       stream, sum.data_ptr<float>(),
-                       sumsq.data_ptr<float>(), g.data_ptr<float>(),
+                       stream, partial_sum.data_ptr<float>(),
+                       partial_ss.data_ptr<float>(), (int)rg.x,
+                       g.data_ptr<float>(),
                        b.data_ptr<float>(),
                        running_mean.data_ptr<float>(),
                        running_var.data_ptr<float>(),
@@ -564,11 +614,16 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd(
   auto stream = at::hip::getCurrentHIPStream();
   auto g = gamma.contiguous().to(at::kFloat);
 
-  auto sum_dy = at::zeros({sh.C}, fopt);
-  auto sum_dy_xhat = at::zeros({sh.C}, fopt);
+  dim3 rg;
+  if (x.scalar_type() == at::kBFloat16)
+    rg = reduce_grid<__hip_bfloat16>(sh.rows, sh.C);
+  else
+    rg = reduce_grid<float>(sh.rows, sh.C);
+  auto sum_dy = at::empty({(int64_t)rg.x, sh.C}, fopt);      // partial_g
+  auto sum_dy_xhat = at::empty({(int64_t)rg.x, sh.C}, fopt); // partial_gx
 #define BN_BRED(RELUF)                                                     \
-  hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, RELUF>),                     \
-                     (reduce_grid<T>(sh.rows, sh.C)), dim3(kBlock), 0,     \
+  hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, RELUF>), rg, dim3(kBlock),  \
+                     0,                                                    \
                      stream, reinterpret_cast<const T*>(x.data_ptr()),     \
                      reinterpret_cast<const T*>(y.data_ptr()),             \
                      reinterpret_cast<const T*>(dy.data_ptr()), sh.rows,   \
@@ -588,7 +643,8 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd(
   int cblocks = (sh.C + kBlock - 1) / kBlock;
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(cblocks), dim3(kBlock), 0,
                      stream, sum_dy.data_ptr<float>(),
-                     sum_dy_xhat.data_ptr<float>(), g.data_ptr<float>(),
+                     sum_dy_xhat.data_ptr<float>(), (int)rg.x,
+                     g.data_ptr<float>(),
                      save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                      dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
                      c_dy.data_ptr<float>(), c_xhat.data_ptr<float>(),
