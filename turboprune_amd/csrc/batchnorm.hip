@@ -318,6 +318,34 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
   if (c0 < C) {
     int64_t r = (int64_t)blockIdx.x * 32 + lane;
     int64_t rstep = (int64_t)gridDim.x * 32;
+    for (; r + rstep < rows; r += 2 * rstep) {
+      int64_t b0 = r * C + c0, b1 = (r + rstep) * C + c0;
+      Octet<T> ox0 = load_octet(x + b0), ox1 = load_octet(x + b1);
+      Octet<T> og0 = load_octet(dy + b0), og1 = load_octet(dy + b1);
+      Octet<T> oy0, oy1;
+      if (RELU) { oy0 = load_octet(y + b0); oy1 = load_octet(y + b1); }
+      f32x2 vx0[P], vg0[P], vy0[P], vx1[P], vg1[P], vy1[P];
+      OctetAcc<T>::unpack(ox0, vx0);
+      OctetAcc<T>::unpack(og0, vg0);
+      OctetAcc<T>::unpack(ox1, vx1);
+      OctetAcc<T>::unpack(og1, vg1);
+      if (RELU) {
+        OctetAcc<T>::unpack(oy0, vy0);
+        OctetAcc<T>::unpack(oy1, vy1);
+      }
+#pragma unroll
+      for (int p = 0; p < P; ++p) {
+        f32x2 g0 = vg0[p], g1 = vg1[p];
+        if (RELU) {
+          g0[0] = vy0[p][0] > 0.f ? g0[0] : 0.f;
+          g0[1] = vy0[p][1] > 0.f ? g0[1] : 0.f;
+          g1[0] = vy1[p][0] > 0.f ? g1[0] : 0.f;
+          g1[1] = vy1[p][1] > 0.f ? g1[1] : 0.f;
+        }
+        s2[p] += g0 + g1;
+        sx2[p] += g0 * vx0[p] + g1 * vx1[p];
+      }
+    }
     for (; r < rows; r += rstep) {
       int64_t base = r * C + c0;
       Octet<T> ox = load_octet(x + base);
