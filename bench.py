@@ -118,16 +118,20 @@ def main():
         sx, sy = next(it)
         static_x = sx.to(memory_format=torch.channels_last).clone()
         static_y = sy.clone()
-        # allocate stable grad buffers; zero_ is captured each replay
-        for _ in range(3):
-            x = static_x
-            opt.zero_grad(set_to_none=False)
-            with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
-                                enabled=bf16):
-                out = model(x)
-                l = TF.cross_entropy(out, static_y)
-            l.backward()
-            opt.step()
+        # warmup on a SIDE stream (capture requirement), stable grad bufs
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(3):
+                opt.zero_grad(set_to_none=False)
+                with torch.autocast(device_type="cuda",
+                                    dtype=torch.bfloat16, enabled=bf16):
+                    out = model(static_x)
+                    l = TF.cross_entropy(out, static_y)
+                l.backward()
+                opt.step()
+        torch.cuda.current_stream().wait_stream(side)
+        del out, l  # drop autograd graph refs before capture
         torch.cuda.synchronize()
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph):

@@ -1,3 +1,5 @@
+import os
+
 import torch
 
 from turboprune_amd.config import compose
@@ -104,3 +106,30 @@ def test_imagenet_loaders_synthetic_fallback(tmp_path):
                            steps_per_epoch=2)
     x, y = next(iter(pair.train_loader))
     assert x.shape == (8, 3, 224, 224)  # 32 total / 4 ranks
+
+
+def test_sharded_imagenet_reader(tmp_path):
+    import subprocess
+    import sys
+    subprocess.run([sys.executable, "scripts/make_shards.py",
+                    "--out", str(tmp_path), "--split", "train",
+                    "--synthetic", "200", "--image-size", "32",
+                    "--shard-size", "64"], check=True,
+                   cwd=os.path.dirname(os.path.dirname(
+                       os.path.abspath(__file__))))
+    from turboprune_amd.data import ShardedImageNet
+    loader = ShardedImageNet(str(tmp_path), "train", batch_size=16,
+                             device=torch.device("cpu"), train=True,
+                             world_size=1, rank=0)
+    batches = list(loader)
+    assert len(batches) >= 10  # 200 imgs, batch 16, drop-last per shard
+    x, y = batches[0]
+    assert x.shape == (16, 3, 32, 32) and x.dtype == torch.float32
+    assert y.dtype == torch.int64
+
+    # rank sharding: 2 ranks see disjoint shard subsets
+    l0 = ShardedImageNet(str(tmp_path), "train", 16, torch.device("cpu"),
+                         True, world_size=2, rank=0)
+    l1 = ShardedImageNet(str(tmp_path), "train", 16, torch.device("cpu"),
+                         True, world_size=2, rank=1)
+    assert set(l0.paths).isdisjoint(set(l1.paths))
