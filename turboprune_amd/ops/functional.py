@@ -14,6 +14,7 @@ kernel so steady-state forwards skip the multiply entirely.
 
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
@@ -76,7 +77,10 @@ def masked_linear(x, weight, mask, bias=None, cache=None, compute_dtype=None):
     if (w.is_cuda and w.dtype == torch.bfloat16 and x.dtype != w.dtype
             and torch.is_autocast_enabled()):
         x = x.to(torch.bfloat16)  # what autocast would do inside F.linear
-    if _backend.use_native(x, w) and x.dim() >= 2:
+    # TURBOPRUNE_GEMM=library routes masked linears through hipBLASLt
+    # instead of the in-house MFMA GEMM (A/B measurement knob)
+    if (os.environ.get("TURBOPRUNE_GEMM", "native") == "native"
+            and _backend.use_native(x, w) and x.dim() >= 2):
         ext = _backend.extension()
         if ext is not None and hasattr(ext, "masked_linear_available") \
                 and ext.masked_linear_available(x, w):
