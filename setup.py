@@ -26,6 +26,7 @@ sources = [
     os.path.join(CSRC, "augment.hip"),
     os.path.join(CSRC, "gemm_masked.hip"),
     os.path.join(CSRC, "batchnorm.hip"),
+    os.path.join(CSRC, "transpose.hip"),
 ]
 
 setup(

@@ -26,6 +26,7 @@ std::tuple<at::Tensor, at::Tensor> linear_bwd(const at::Tensor&,
                                               const at::Tensor&);
 at::Tensor gemm_bf16(const at::Tensor&, const at::Tensor&, bool, bool);
 // fused NHWC batchnorm (batchnorm.hip)
+at::Tensor transpose2d(const at::Tensor&);
 bool bn_fast_path_ok(const at::Tensor&);
 std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd(
     const at::Tensor&, const c10::optional<at::Tensor>&, const at::Tensor&,
@@ -61,6 +62,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA GEMM grads (grad_x, grad_w)");
   m.def("gemm_bf16", &turboprune::gemm_bf16,
         "raw MFMA bf16 GEMM (testing entry)");
+  m.def("transpose2d", &turboprune::transpose2d,
+        "tiled LDS 2-D transpose (2-byte dtypes)");
   m.def("bn_fast_path_ok", &turboprune::bn_fast_path_ok);
   m.def("bn_fwd", &turboprune::bn_fwd,
         "fused NHWC batchnorm(+relu)(+residual) forward");

@@ -30,6 +30,8 @@
 
 namespace turboprune {
 
+at::Tensor transpose2d(const at::Tensor&);  // transpose.hip
+
 using bf16x8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
@@ -147,20 +149,34 @@ __global__ void gemm_bt_kernel(const __hip_bfloat16* __restrict__ A,
     cur ^= 1;
   }
 
-  // ---- epilogue: C[row][col] (+bias) ----------------------------------
+  // ---- epilogue: stage the C tile in LDS, then coalesced 16B row
+  // stores (the MFMA C fragment is column-per-lane: direct stores are
+  // 2-byte scattered — issue-bound AND line-wasteful, guide T21) -------
+  __syncthreads();  // all waves done reading sA/sB
+  OutT* cs = reinterpret_cast<OutT*>(smem);  // 128x128 OutT fits 64 KiB
 #pragma unroll
   for (int mi = 0; mi < MREP; ++mi) {
 #pragma unroll
     for (int ni = 0; ni < NREP; ++ni) {
-      int col = tile_n + wc * WN + ni * 16 + (lane & 15);
+      int col = wc * WN + ni * 16 + (lane & 15);
 #pragma unroll
       for (int j = 0; j < 4; ++j) {
-        int row = tile_m + wr * WM + mi * 16 + (lane >> 4) * 4 + j;
+        int row = wr * WM + mi * 16 + (lane >> 4) * 4 + j;
         float v = acc[mi][ni][j];
-        if (HAS_BIAS) v += bias[col];
-        C[(int64_t)row * N + col] = from_float<OutT>(v);
+        if (HAS_BIAS) v += bias[tile_n + col];
+        cs[row * BN + col] = from_float<OutT>(v);
       }
     }
+  }
+  __syncthreads();
+  constexpr int EV = 16 / sizeof(OutT);  // elems per 16B chunk
+  int chunks_per_row = BN / EV;
+  int total_chunks = BM * chunks_per_row;
+  for (int idx = threadIdx.x; idx < total_chunks; idx += blockDim.x) {
+    int r = idx / chunks_per_row;
+    int cc = (idx % chunks_per_row) * EV;
+    *reinterpret_cast<uint4*>(&C[(int64_t)(tile_m + r) * N + tile_n + cc]) =
+        *reinterpret_cast<const uint4*>(&cs[r * BN + cc]);
   }
 }
 
@@ -261,11 +277,13 @@ std::tuple<at::Tensor, at::Tensor> linear_bwd(const at::Tensor& grad_y,
   int64_t K = x.size(-1);
   auto gy2 = grad_y.reshape({M, N}).contiguous();
   auto x2 = x.reshape({M, K}).contiguous();
+  // transposes via the tiled LDS transpose kernel (eager .t().contiguous()
+  // measured ~1.1 TB/s and dominated the DeiT step)
   // grad_x (M,K) = gy (M,N) @ w (N,K): B_t = w^T (K,N) K-major in N
-  auto wT = w.t().contiguous();
+  auto wT = transpose2d(w.contiguous());
   auto gx = gemm_bt(gy2, wT, c10::nullopt, false);
   // grad_w (N,K) = gy^T (N,M) @ x (M,K): A = gy^T, B_t = x^T (K,M)
-  auto gw = gemm_bt(gy2.t().contiguous(), x2.t().contiguous(), c10::nullopt,
+  auto gw = gemm_bt(transpose2d(gy2), transpose2d(x2), c10::nullopt,
                     false);
   auto x_sizes = x.sizes().vec();
   return {gx.reshape(x_sizes), gw};
