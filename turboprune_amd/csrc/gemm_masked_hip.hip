@@ -49,6 +49,10 @@ TP_DEVICE int lds_byte(int row, int k) {
   return row * (BK * 2) + blk * 16 + (k & 7) * 2;
 }
 
+// gridDim.y > 1 = split-K: slab blockIdx.y computes K-tiles
+// [y*per, min((y+1)*per, K/BK)) into C + y*M*N (fp32, no bias); a reduce
+// kernel sums the slabs. Small-output deep-K GEMMs (grad_w: e.g. DeiT
+// 1152x384 over K=50k = 27 tiles) otherwise strand 90% of the CUs.
 template <typename OutT, bool HAS_BIAS>
 __launch_bounds__(256)
 __global__ void gemm_bt_kernel(const __hip_bfloat16* __restrict__ A,
@@ -113,13 +117,18 @@ __global__ void gemm_bt_kernel(const __hip_bfloat16* __restrict__ A,
 
   f32x4 acc[MREP][NREP] = {};
 
-  int nt = K / BK;
-  stage(0, 0);
+  int total_kt = K / BK;
+  int per = (total_kt + gridDim.y - 1) / gridDim.y;
+  int kt0 = blockIdx.y * per;
+  int kt1 = min(kt0 + per, total_kt);
+  C += (int64_t)blockIdx.y * M * N;
+
+  if (kt0 < kt1) stage(0, kt0);  // empty tail slab writes zeros
   __syncthreads();  // carries vmcnt(0): glds drained
 
   int cur = 0;
-  for (int t = 0; t < nt; ++t) {
-    if (t + 1 < nt) stage(cur ^ 1, t + 1);
+  for (int t = kt0; t < kt1; ++t) {
+    if (t + 1 < kt1) stage(cur ^ 1, t + 1);
 
     // compute on buf `cur`: 2 k-steps of 32
 #pragma unroll
@@ -181,6 +190,22 @@ __global__ void gemm_bt_kernel(const __hip_bfloat16* __restrict__ A,
   }
 }
 
+// split-K slab reduction: out[i] = sum_s partial[s][i] (+bias) -> OutT
+template <typename OutT, bool HAS_BIAS>
+__global__ void splitk_reduce_kernel(const float* __restrict__ partial,
+                                     OutT* __restrict__ out,
+                                     const float* __restrict__ bias,
+                                     int64_t mn, int N, int slabs) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < mn;
+       i += stride) {
+    float v = 0.f;
+    for (int s = 0; s < slabs; ++s) v += partial[(int64_t)s * mn + i];
+    if (HAS_BIAS) v += bias[i % N];
+    out[i] = from_float<OutT>(v);
+  }
+}
+
 // ---------------------------------------------------------------- host
 static at::Tensor pad_to(const at::Tensor& t, int64_t r, int64_t c) {
   if (t.size(0) == r && t.size(1) == c) return t.contiguous();
@@ -217,24 +242,55 @@ at::Tensor gemm_bt(const at::Tensor& A, const at::Tensor& B,
                        A.options().dtype(out_fp32 ? at::kFloat
                                                   : at::kBFloat16));
   int grid_m = Mp / BM, grid_n = Np / BN;
-  dim3 grid(grid_m * grid_n);
+  int tiles = grid_m * grid_n;
+  int total_kt = (int)(Kp / BK);
+  // split-K when the tile grid underfills the 256-CU chip and K is deep
+  int splitk = 1;
+  while (tiles * splitk < 384 && splitk * 2 * 2 <= total_kt &&
+         splitk < 32)
+    splitk *= 2;
   auto stream = at::hip::getCurrentHIPStream();
   auto* ap = reinterpret_cast<const __hip_bfloat16*>(Ap.data_ptr());
   auto* bp = reinterpret_cast<const __hip_bfloat16*>(Bp.data_ptr());
   const float* biasptr = has_bias ? biasp.data_ptr<float>() : nullptr;
 
-#define TP_GEMM(OutT, HB)                                                  \
-  hipLaunchKernelGGL((gemm_bt_kernel<OutT, HB>), grid, dim3(256), 0,       \
-                     stream, ap, bp,                                       \
-                     reinterpret_cast<OutT*>(out.data_ptr()), biasptr,     \
-                     (int)Mp, (int)Np, (int)Kp, grid_n)
-  if (out_fp32) {
-    if (has_bias) TP_GEMM(float, true); else TP_GEMM(float, false);
+  if (splitk > 1) {
+    auto partial = at::empty({splitk, Mp, Np},
+                             A.options().dtype(at::kFloat));
+    dim3 grid(tiles, splitk);
+    hipLaunchKernelGGL((gemm_bt_kernel<float, false>), grid, dim3(256), 0,
+                       stream, ap, bp, partial.data_ptr<float>(), nullptr,
+                       (int)Mp, (int)Np, (int)Kp, grid_n);
+    int64_t mn = Mp * Np;
+    int rgrid = elementwise_grid(mn, kBlock, 4);
+#define TP_RED(OutT, HB)                                                   \
+    hipLaunchKernelGGL((splitk_reduce_kernel<OutT, HB>), dim3(rgrid),      \
+                       dim3(kBlock), 0, stream,                            \
+                       partial.data_ptr<float>(),                          \
+                       reinterpret_cast<OutT*>(out.data_ptr()), biasptr,   \
+                       mn, (int)Np, splitk)
+    if (out_fp32) {
+      if (has_bias) TP_RED(float, true); else TP_RED(float, false);
+    } else {
+      if (has_bias) TP_RED(__hip_bfloat16, true);
+      else TP_RED(__hip_bfloat16, false);
+    }
+#undef TP_RED
   } else {
-    if (has_bias) TP_GEMM(__hip_bfloat16, true);
-    else TP_GEMM(__hip_bfloat16, false);
-  }
+    dim3 grid(tiles, 1);
+#define TP_GEMM(OutT, HB)                                                  \
+    hipLaunchKernelGGL((gemm_bt_kernel<OutT, HB>), grid, dim3(256), 0,     \
+                       stream, ap, bp,                                     \
+                       reinterpret_cast<OutT*>(out.data_ptr()), biasptr,   \
+                       (int)Mp, (int)Np, (int)Kp, grid_n)
+    if (out_fp32) {
+      if (has_bias) TP_GEMM(float, true); else TP_GEMM(float, false);
+    } else {
+      if (has_bias) TP_GEMM(__hip_bfloat16, true);
+      else TP_GEMM(__hip_bfloat16, false);
+    }
 #undef TP_GEMM
+  }
   if (Mp != M || Np != N)
     return out.narrow(0, 0, M).narrow(1, 0, N).contiguous();
   return out;
