@@ -69,6 +69,17 @@ class BaseHarness:
             # bf16 masked-weight caches, maintained by the fused optimizer
             self.model.enable_caches(self.amp_dtype)
 
+        if bool(cfg.select("model_params.use_compile", False)):
+            # reference surface only (standard_pruning_harness.py:141-142,
+            # off in every shipped config). torch.compile on ROCm routes
+            # through Triton, which this MI355X-native build deliberately
+            # excludes — the hot path is hand-written HIP + hipGraphs.
+            import warnings
+            warnings.warn("model_params.use_compile is accepted for config "
+                          "compatibility but ignored: this build uses "
+                          "hand-written HIP kernels (+ hipGraph capture) "
+                          "instead of torch.compile/Triton")
+
         self.ddp_model: torch.nn.Module = (
             P.wrap_ddp(self.model, cfg, self.device)
             if self.distributed else self.model)
