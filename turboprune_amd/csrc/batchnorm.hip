@@ -255,10 +255,13 @@ __global__ void bn_eval_coeffs_kernel(const float* __restrict__ running_mean,
 
 // ---------------- forward apply ------------------------------------------
 // flat grid-stride over 16B octets; scale/shift cached in LDS (C <= 4096)
-template <typename T, bool RELU, bool RES>
+// WM: also emit a per-octet ReLU bitmask (bit j = element j survived) so
+// the backward never re-reads y (1/16 of the bytes).
+template <typename T, bool RELU, bool RES, bool WM>
 __global__ void bn_apply_kernel(const T* __restrict__ x,
                                 const T* __restrict__ res,
                                 T* __restrict__ y,
+                                uint8_t* __restrict__ rmask,
                                 const float* __restrict__ scale,
                                 const float* __restrict__ shift,
                                 int64_t rows, int C) {
@@ -282,14 +285,19 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
     Octet<T> r;
     if (RES) r = load_octet(res + base);
     Octet<T> out;
+    unsigned mbits = 0;
 #pragma unroll
     for (int j = 0; j < VN; ++j) {
       float v = o.get(j) * lscale[c0 + j] + lshift[c0 + j];
       if (RES) v += r.get(j);
-      if (RELU) v = fmaxf(v, 0.f);
+      if (RELU) {
+        if (WM && v > 0.f) mbits |= (1u << j);
+        v = fmaxf(v, 0.f);
+      }
       out.set(j, v);
     }
     store_octet(y + base, out);
+    if (RELU && WM) rmask[i] = (uint8_t)mbits;
   }
 }
 
@@ -299,7 +307,7 @@ __global__ void bn_apply_kernel(const T* __restrict__ x,
 // in the finalize kernel — keeps the hot loop at ~3 packed VALU/dword).
 template <typename T, bool RELU>
 __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
-                                     const T* __restrict__ y,
+                                     const uint8_t* __restrict__ rmask,
                                      const T* __restrict__ dy, int64_t rows,
                                      int C,
                                      float* __restrict__ partial_g,
@@ -318,29 +326,29 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
   if (c0 < C) {
     int64_t r = (int64_t)blockIdx.x * 32 + lane;
     int64_t rstep = (int64_t)gridDim.x * 32;
+    int n_oct = C / VN;
     for (; r + rstep < rows; r += 2 * rstep) {
       int64_t b0 = r * C + c0, b1 = (r + rstep) * C + c0;
       Octet<T> ox0 = load_octet(x + b0), ox1 = load_octet(x + b1);
       Octet<T> og0 = load_octet(dy + b0), og1 = load_octet(dy + b1);
-      Octet<T> oy0, oy1;
-      if (RELU) { oy0 = load_octet(y + b0); oy1 = load_octet(y + b1); }
-      f32x2 vx0[P], vg0[P], vy0[P], vx1[P], vg1[P], vy1[P];
+      unsigned m0 = 0xffu, m1 = 0xffu;
+      if (RELU) {
+        m0 = rmask[r * n_oct + oct];
+        m1 = rmask[(r + rstep) * n_oct + oct];
+      }
+      f32x2 vx0[P], vg0[P], vx1[P], vg1[P];
       OctetAcc<T>::unpack(ox0, vx0);
       OctetAcc<T>::unpack(og0, vg0);
       OctetAcc<T>::unpack(ox1, vx1);
       OctetAcc<T>::unpack(og1, vg1);
-      if (RELU) {
-        OctetAcc<T>::unpack(oy0, vy0);
-        OctetAcc<T>::unpack(oy1, vy1);
-      }
 #pragma unroll
       for (int p = 0; p < P; ++p) {
         f32x2 g0 = vg0[p], g1 = vg1[p];
         if (RELU) {
-          g0[0] = vy0[p][0] > 0.f ? g0[0] : 0.f;
-          g0[1] = vy0[p][1] > 0.f ? g0[1] : 0.f;
-          g1[0] = vy1[p][0] > 0.f ? g1[0] : 0.f;
-          g1[1] = vy1[p][1] > 0.f ? g1[1] : 0.f;
+          g0[0] = (m0 >> (2 * p)) & 1 ? g0[0] : 0.f;
+          g0[1] = (m0 >> (2 * p + 1)) & 1 ? g0[1] : 0.f;
+          g1[0] = (m1 >> (2 * p)) & 1 ? g1[0] : 0.f;
+          g1[1] = (m1 >> (2 * p + 1)) & 1 ? g1[1] : 0.f;
         }
         s2[p] += g0 + g1;
         sx2[p] += g0 * vx0[p] + g1 * vx1[p];
@@ -350,18 +358,16 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
       int64_t base = r * C + c0;
       Octet<T> ox = load_octet(x + base);
       Octet<T> og = load_octet(dy + base);
-      Octet<T> oy;
-      if (RELU) oy = load_octet(y + base);
-      f32x2 vx[P], vg[P], vy[P];
+      unsigned mb = RELU ? rmask[r * n_oct + oct] : 0xffu;
+      f32x2 vx[P], vg[P];
       OctetAcc<T>::unpack(ox, vx);
       OctetAcc<T>::unpack(og, vg);
-      if (RELU) OctetAcc<T>::unpack(oy, vy);
 #pragma unroll
       for (int p = 0; p < P; ++p) {
         f32x2 g = vg[p];
         if (RELU) {
-          g[0] = vy[p][0] > 0.f ? g[0] : 0.f;
-          g[1] = vy[p][1] > 0.f ? g[1] : 0.f;
+          g[0] = (mb >> (2 * p)) & 1 ? g[0] : 0.f;
+          g[1] = (mb >> (2 * p + 1)) & 1 ? g[1] : 0.f;
         }
         s2[p] += g;
         sx2[p] += g * vx[p];
@@ -449,7 +455,7 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ partial_g,
 // LDS caches 5 coeff arrays: mean, rstd, c_dy, c_xhat, c_const
 template <typename T, bool RELU, bool RES>
 __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
-                                    const T* __restrict__ y,
+                                    const uint8_t* __restrict__ rmask,
                                     const T* __restrict__ dy,
                                     T* __restrict__ dx,
                                     T* __restrict__ dres,
@@ -483,14 +489,13 @@ __global__ void bn_bwd_apply_kernel(const T* __restrict__ x,
     int64_t base = (i / n_oct) * C + c0;
     Octet<T> ox = load_octet(x + base);
     Octet<T> og = load_octet(dy + base);
-    Octet<T> oy;
-    if (RELU) oy = load_octet(y + base);
+    unsigned mb = RELU ? rmask[i] : 0xffu;
     Octet<T> odx, odr;
 #pragma unroll
     for (int j = 0; j < VN; ++j) {
       int c = c0 + j;
       float g = og.get(j);
-      if (RELU && oy.get(j) <= 0.f) g = 0.f;
+      if (RELU && !((mb >> j) & 1)) g = 0.f;
       if (RES) odr.set(j, g);
       float xhat = (ox.get(j) - lmean[c]) * lrstd[c];
       odx.set(j, lcdy[c] * g + lcx[c] * xhat + lcc[c]);
@@ -537,8 +542,8 @@ bool bn_fast_path_ok(const at::Tensor& x) {
   return C % vn == 0 && C <= 4096;
 }
 
-// returns (y, save_mean, save_rstd)
-std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd(
+// returns (y, save_mean, save_rstd, relu_mask-or-undef)
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_fwd(
     const at::Tensor& x, const c10::optional<at::Tensor>& residual,
     const at::Tensor& gamma, const at::Tensor& beta,
     at::Tensor running_mean, at::Tensor running_var, bool training,
@@ -610,31 +615,53 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd(
     res = residual->contiguous(at::MemoryFormat::ChannelsLast);
     TORCH_CHECK(res.sizes() == x.sizes());
   }
+  // ReLU bitmask (1 bit/elem, 1 byte/octet): written in training so the
+  // backward never re-reads y
+  int vn = x.scalar_type() == at::kBFloat16 ? 8 : 4;
+  at::Tensor rmask;
+  bool write_mask = relu && training;
+  uint8_t* rmask_ptr = nullptr;
+  if (write_mask) {
+    rmask = at::empty({sh.rows * (sh.C / vn)}, x.options().dtype(at::kByte));
+    rmask_ptr = rmask.data_ptr<uint8_t>();
+  }
   size_t lds_bytes = 2 * sh.C * sizeof(float);
-#define BN_APPLY(RELUF, RESF)                                             \
+#define BN_APPLY(RELUF, RESF, WMF)                                        \
   hipLaunchKernelGGL(                                                     \
-      (bn_apply_kernel<T, RELUF, RESF>),                                  \
+      (bn_apply_kernel<T, RELUF, RESF, WMF>),                             \
       dim3(apply_grid(sh.rows, sh.C, Octet<T>::kN)), dim3(kBlock),        \
       lds_bytes, stream, reinterpret_cast<const T*>(x.data_ptr()),        \
       RESF ? reinterpret_cast<const T*>(res.data_ptr()) : nullptr,        \
-      reinterpret_cast<T*>(y.data_ptr()), scale.data_ptr<float>(),        \
+      reinterpret_cast<T*>(y.data_ptr()), rmask_ptr,                      \
+      scale.data_ptr<float>(),                                            \
       shift.data_ptr<float>(), sh.rows, sh.C)
 #define BN_APPLY_D()                                                      \
-  if (relu) { if (has_res) BN_APPLY(true, true);                          \
-              else BN_APPLY(true, false); }                               \
-  else { if (has_res) BN_APPLY(false, true);                              \
-         else BN_APPLY(false, false); }
+  if (relu && write_mask) {                                               \
+    if (has_res) BN_APPLY(true, true, true);                              \
+    else BN_APPLY(true, false, true);                                     \
+  } else if (relu) {                                                      \
+    if (has_res) BN_APPLY(true, true, false);                             \
+    else BN_APPLY(true, false, false);                                    \
+  } else {                                                                \
+    if (has_res) BN_APPLY(false, true, false);                            \
+    else BN_APPLY(false, false, false);                                   \
+  }
   BN_DISPATCH_T(BN_APPLY_D)
 #undef BN_APPLY_D
 #undef BN_APPLY
-  return {y, save_mean, save_rstd};
+  return {y, save_mean, save_rstd, rmask};
 }
 
 // returns (dx, dgamma, dbeta, dres?) — dres empty when no residual
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd(
-    const at::Tensor& x, const at::Tensor& y, const at::Tensor& dy_in,
+    const at::Tensor& x, const at::Tensor& relu_mask,
+    const at::Tensor& dy_in,
     const at::Tensor& gamma, const at::Tensor& save_mean,
     const at::Tensor& save_rstd, bool relu, bool has_res) {
+  TORCH_CHECK(!relu || relu_mask.defined(),
+              "bn_bwd: relu requires the forward's relu_mask");
+  const uint8_t* rmask_ptr =
+      relu ? relu_mask.data_ptr<uint8_t>() : nullptr;
   auto sh = bn_shape(x);
   auto dy = dy_in.contiguous(at::MemoryFormat::ChannelsLast);
   auto fopt = x.options().dtype(at::kFloat);
@@ -652,7 +679,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd(
   hipLaunchKernelGGL((bn_bwd_reduce_kernel<T, RELUF>), rg, dim3(kBlock),  \
                      0,                                                    \
                      stream, reinterpret_cast<const T*>(x.data_ptr()),     \
-                     reinterpret_cast<const T*>(y.data_ptr()),             \
+                     rmask_ptr,                                            \
                      reinterpret_cast<const T*>(dy.data_ptr()), sh.rows,   \
                      sh.C, sum_dy.data_ptr<float>(),                       \
                      sum_dy_xhat.data_ptr<float>())
@@ -687,7 +714,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd(
       (bn_bwd_apply_kernel<T, RELUF, RESF>),                               \
       dim3(apply_grid(sh.rows, sh.C, Octet<T>::kN)), dim3(kBlock),         \
       lds_bytes, stream, reinterpret_cast<const T*>(x.data_ptr()),         \
-      reinterpret_cast<const T*>(y.data_ptr()),                            \
+      rmask_ptr,                                                           \
       reinterpret_cast<const T*>(dy.data_ptr()),                           \
       reinterpret_cast<T*>(dx.data_ptr()),                                 \
       RESF ? reinterpret_cast<T*>(dres.data_ptr()) : nullptr,              \

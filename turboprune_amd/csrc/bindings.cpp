@@ -28,7 +28,7 @@ at::Tensor gemm_bf16(const at::Tensor&, const at::Tensor&, bool, bool);
 // fused NHWC batchnorm (batchnorm.hip)
 at::Tensor transpose2d(const at::Tensor&);
 bool bn_fast_path_ok(const at::Tensor&);
-std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd(
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_fwd(
     const at::Tensor&, const c10::optional<at::Tensor>&, const at::Tensor&,
     const at::Tensor&, at::Tensor, at::Tensor, bool, double, double, bool);
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd(

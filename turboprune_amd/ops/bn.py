@@ -30,20 +30,20 @@ class _FusedBN(torch.autograd.Function):
     def forward(ctx, x, residual, gamma, beta, running_mean, running_var,
                 training, momentum, eps, relu):
         ext = _backend.extension()
-        y, mean, rstd = ext.bn_fwd(x, residual, gamma, beta, running_mean,
-                                   running_var, training, momentum, eps,
-                                   relu)
-        ctx.save_for_backward(x, y, gamma, mean, rstd)
+        y, mean, rstd, rmask = ext.bn_fwd(x, residual, gamma, beta,
+                                          running_mean, running_var,
+                                          training, momentum, eps, relu)
+        ctx.save_for_backward(x, gamma, mean, rstd, rmask)
         ctx.relu = relu
         ctx.has_res = residual is not None
         return y
 
     @staticmethod
     def backward(ctx, dy):
-        x, y, gamma, mean, rstd = ctx.saved_tensors
+        x, gamma, mean, rstd, rmask = ctx.saved_tensors
         ext = _backend.extension()
-        dx, dgamma, dbeta, dres = ext.bn_bwd(x, y, dy, gamma, mean, rstd,
-                                             ctx.relu, ctx.has_res)
+        dx, dgamma, dbeta, dres = ext.bn_bwd(x, rmask, dy, gamma, mean,
+                                             rstd, ctx.relu, ctx.has_res)
         return (dx, dres if ctx.has_res else None,
                 dgamma.to(gamma.dtype), dbeta.to(gamma.dtype),
                 None, None, None, None, None, None)
@@ -90,9 +90,9 @@ def bn_act(bn: nn.BatchNorm2d, x: torch.Tensor,
                               momentum, bn.eps, relu)
     ext = _backend.extension()
     with torch.no_grad():
-        y, _, _ = ext.bn_fwd(x, residual, bn.weight, bn.bias,
-                             bn.running_mean, bn.running_var, bn.training,
-                             momentum, bn.eps, relu)
+        y = ext.bn_fwd(x, residual, bn.weight, bn.bias,
+                       bn.running_mean, bn.running_var, bn.training,
+                       momentum, bn.eps, relu)[0]
     return y
 
 
