@@ -46,10 +46,10 @@ def main():
             return ext.bn_fwd(x, res, gamma, beta, rm, rv, True, 0.1,
                               1e-5, True)
 
-        y, mean, rstd = fused_fwd()
+        y, mean, rstd, rmask = fused_fwd()
 
         def fused_bwd():
-            return ext.bn_bwd(x, y, dy, gamma, mean, rstd, True, True)
+            return ext.bn_bwd(x, rmask, dy, gamma, mean, rstd, True, True)
 
         def miopen_fwd():
             out = torch.nn.functional.batch_norm(

@@ -42,6 +42,8 @@ class _FusedBN(torch.autograd.Function):
     def backward(ctx, dy):
         x, gamma, mean, rstd, rmask = ctx.saved_tensors
         ext = _backend.extension()
+        if rmask is None:  # relu=False path: no bitmask was written
+            rmask = torch.Tensor()
         dx, dgamma, dbeta, dres = ext.bn_bwd(x, rmask, dy, gamma, mean,
                                              rstd, ctx.relu, ctx.has_res)
         return (dx, dres if ctx.has_res else None,
