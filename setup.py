@@ -27,6 +27,7 @@ sources = [
     os.path.join(CSRC, "gemm_masked.hip"),
     os.path.join(CSRC, "batchnorm.hip"),
     os.path.join(CSRC, "transpose.hip"),
+    os.path.join(CSRC, "maxpool.hip"),
 ]
 
 setup(

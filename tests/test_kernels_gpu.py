@@ -256,3 +256,25 @@ def test_masked_linear_uses_gemm(ext):
         x.float(), (layer.weight * layer.mask).float(),
         layer.bias.float())
     assert (y.float() - ref).abs().max().item() < 0.5
+
+
+# ---------------------------------------------------------------- maxpool
+@pytest.mark.parametrize("shape,k,s,p", [
+    ((4, 64, 32, 32), 3, 2, 1),    # resnet stem
+    ((2, 128, 16, 16), 2, 2, 0),   # vgg
+    ((2, 64, 15, 15), 3, 2, 1),    # odd spatial
+])
+def test_fused_maxpool_matches_torch(ext, shape, k, s, p):
+    torch.manual_seed(0)
+    from turboprune_amd.ops.pool import FusedMaxPool2d
+    x = torch.randn(*shape, device=DEV).to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last).requires_grad_()
+    x2 = x.detach().clone().requires_grad_()
+    pool = FusedMaxPool2d(k, s, p)
+    y = pool(x)
+    y_ref = torch.nn.functional.max_pool2d(x2, k, s, p)
+    assert torch.equal(y.float(), y_ref.float())
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    y_ref.backward(dy)
+    assert torch.allclose(x.grad.float(), x2.grad.float(), atol=1e-2)

@@ -27,6 +27,10 @@ std::tuple<at::Tensor, at::Tensor> linear_bwd(const at::Tensor&,
 at::Tensor gemm_bf16(const at::Tensor&, const at::Tensor&, bool, bool);
 // fused NHWC batchnorm (batchnorm.hip)
 at::Tensor transpose2d(const at::Tensor&);
+std::tuple<at::Tensor, at::Tensor> maxpool_fwd(const at::Tensor&, int, int,
+                                               int, int, int, int);
+at::Tensor maxpool_bwd(const at::Tensor&, const at::Tensor&, int, int, int,
+                       int, int, int, int, int, int, int);
 bool bn_fast_path_ok(const at::Tensor&);
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_fwd(
     const at::Tensor&, const c10::optional<at::Tensor>&, const at::Tensor&,
@@ -62,6 +66,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA GEMM grads (grad_x, grad_w)");
   m.def("gemm_bf16", &turboprune::gemm_bf16,
         "raw MFMA bf16 GEMM (testing entry)");
+  m.def("maxpool_fwd", &turboprune::maxpool_fwd,
+        "NHWC maxpool forward -> (y, argmax bytes)");
+  m.def("maxpool_bwd", &turboprune::maxpool_bwd,
+        "NHWC maxpool backward (gather)");
   m.def("transpose2d", &turboprune::transpose2d,
         "tiled LDS 2-D transpose (2-byte dtypes)");
   m.def("bn_fast_path_ok", &turboprune::bn_fast_path_ok);

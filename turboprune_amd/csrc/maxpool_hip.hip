@@ -1,0 +1,213 @@
+#include "hip/hip_runtime.h"
+// NHWC MaxPool2d forward/backward (ResNet stem 3x3s2p1, VGG 2x2s2).
+//
+// Forward stores a per-element uint8 window-argmax index; backward is a
+// GATHER over the <=4 covering windows (no atomics). Channel octets are
+// 16-byte vectors; eager max_pool2d on channels_last bf16 measured
+// 524/1231 us fwd/bwd per ResNet50 step.
+
+#include <ATen/ATen.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace turboprune {
+
+template <typename T>
+struct PoolVec;
+template <>
+struct PoolVec<__hip_bfloat16> {
+  static constexpr int kN = 8;
+  using IdxStore = unsigned long long;  // 8 bytes
+};
+template <>
+struct PoolVec<float> {
+  static constexpr int kN = 4;
+  using IdxStore = unsigned int;  // 4 bytes
+};
+
+template <typename T>
+__global__ void maxpool_fwd_kernel(const T* __restrict__ x,
+                                   T* __restrict__ y,
+                                   uint8_t* __restrict__ idx, int N, int C,
+                                   int Hi, int Wi, int Ho, int Wo, int kh,
+                                   int kw, int sh, int sw, int ph, int pw) {
+  constexpr int VN = PoolVec<T>::kN;
+  int n_oct = C / VN;
+  int64_t total = (int64_t)N * Ho * Wo * n_oct;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    int oct = (int)(i % n_oct);
+    int64_t row = i / n_oct;
+    int wo = (int)(row % Wo);
+    int64_t row2 = row / Wo;
+    int ho = (int)(row2 % Ho);
+    int n = (int)(row2 / Ho);
+    int c0 = oct * VN;
+
+    float best[VN];
+    int bidx[VN];
+#pragma unroll
+    for (int j = 0; j < VN; ++j) { best[j] = -INFINITY; bidx[j] = 0; }
+    int h0 = ho * sh - ph, w0 = wo * sw - pw;
+    for (int dh = 0; dh < kh; ++dh) {
+      int hi = h0 + dh;
+      if (hi < 0 || hi >= Hi) continue;
+      for (int dw = 0; dw < kw; ++dw) {
+        int wi = w0 + dw;
+        if (wi < 0 || wi >= Wi) continue;
+        const T* src = x + (((int64_t)n * Hi + hi) * Wi + wi) * C + c0;
+        uint4 raw = *reinterpret_cast<const uint4*>(src);
+        const T* v = reinterpret_cast<const T*>(&raw);
+        int pos = dh * kw + dw;
+#pragma unroll
+        for (int j = 0; j < VN; ++j) {
+          float f = to_float<T>(v[j]);
+          if (f > best[j]) { best[j] = f; bidx[j] = pos; }
+        }
+      }
+    }
+    T outv[VN];
+    uint8_t outi[VN];
+#pragma unroll
+    for (int j = 0; j < VN; ++j) {
+      outv[j] = from_float<T>(best[j]);
+      outi[j] = (uint8_t)bidx[j];
+    }
+    int64_t obase = row * C + c0;
+    *reinterpret_cast<uint4*>(y + obase) =
+        *reinterpret_cast<const uint4*>(&outv[0]);
+    *reinterpret_cast<typename PoolVec<T>::IdxStore*>(idx + obase) =
+        *reinterpret_cast<const typename PoolVec<T>::IdxStore*>(&outi[0]);
+  }
+}
+
+template <typename T>
+__global__ void maxpool_bwd_kernel(const T* __restrict__ dy,
+                                   const uint8_t* __restrict__ idx,
+                                   T* __restrict__ dx, int N, int C, int Hi,
+                                   int Wi, int Ho, int Wo, int kh, int kw,
+                                   int sh, int sw, int ph, int pw) {
+  constexpr int VN = PoolVec<T>::kN;
+  int n_oct = C / VN;
+  int64_t total = (int64_t)N * Hi * Wi * n_oct;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < total;
+       i += stride) {
+    int oct = (int)(i % n_oct);
+    int64_t row = i / n_oct;
+    int wi = (int)(row % Wi);
+    int64_t row2 = row / Wi;
+    int hi = (int)(row2 % Hi);
+    int n = (int)(row2 / Hi);
+    int c0 = oct * VN;
+
+    float acc[VN];
+#pragma unroll
+    for (int j = 0; j < VN; ++j) acc[j] = 0.f;
+
+    // output windows covering (hi, wi): ho*sh - ph <= hi < ho*sh - ph + kh
+    int ho_lo = (hi + ph - kh + sh) / sh;  // ceil((hi+ph-kh+1)/sh)
+    if (ho_lo < 0) ho_lo = 0;
+    int ho_hi = (hi + ph) / sh;
+    if (ho_hi >= Ho) ho_hi = Ho - 1;
+    int wo_lo = (wi + pw - kw + sw) / sw;
+    if (wo_lo < 0) wo_lo = 0;
+    int wo_hi = (wi + pw) / sw;
+    if (wo_hi >= Wo) wo_hi = Wo - 1;
+
+    for (int ho = ho_lo; ho <= ho_hi; ++ho) {
+      int dh = hi - (ho * sh - ph);
+      if (dh < 0 || dh >= kh) continue;
+      for (int wo = wo_lo; wo <= wo_hi; ++wo) {
+        int dw = wi - (wo * sw - pw);
+        if (dw < 0 || dw >= kw) continue;
+        int pos = dh * kw + dw;
+        int64_t obase = (((int64_t)n * Ho + ho) * Wo + wo) * C + c0;
+        uint4 graw = *reinterpret_cast<const uint4*>(dy + obase);
+        const T* g = reinterpret_cast<const T*>(&graw);
+        typename PoolVec<T>::IdxStore iraw =
+            *reinterpret_cast<const typename PoolVec<T>::IdxStore*>(
+                idx + obase);
+        const uint8_t* ip = reinterpret_cast<const uint8_t*>(&iraw);
+#pragma unroll
+        for (int j = 0; j < VN; ++j)
+          if (ip[j] == pos) acc[j] += to_float<T>(g[j]);
+      }
+    }
+    T outv[VN];
+#pragma unroll
+    for (int j = 0; j < VN; ++j) outv[j] = from_float<T>(acc[j]);
+    *reinterpret_cast<uint4*>(dx + row * C + c0) =
+        *reinterpret_cast<const uint4*>(&outv[0]);
+  }
+}
+
+static int pool_out(int in, int k, int s, int p) {
+  return (in + 2 * p - k) / s + 1;
+}
+
+std::tuple<at::Tensor, at::Tensor> maxpool_fwd(const at::Tensor& x, int kh,
+                                               int kw, int sh, int sw,
+                                               int ph, int pw) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 &&
+              x.is_contiguous(at::MemoryFormat::ChannelsLast));
+  int N = x.size(0), C = x.size(1), Hi = x.size(2), Wi = x.size(3);
+  int vn = x.scalar_type() == at::kBFloat16 ? 8 : 4;
+  TORCH_CHECK(C % vn == 0, "maxpool: C must be a multiple of ", vn);
+  TORCH_CHECK(kh * kw <= 255);
+  int Ho = pool_out(Hi, kh, sh, ph), Wo = pool_out(Wi, kw, sw, pw);
+  auto y = at::empty({N, C, Ho, Wo},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  auto idx = at::empty({(int64_t)N * Ho * Wo * C},
+                       x.options().dtype(at::kByte));
+  int64_t total = (int64_t)N * Ho * Wo * (C / vn);
+  int grid = elementwise_grid(total, kBlock, 2);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (x.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(maxpool_fwd_kernel<__hip_bfloat16>, dim3(grid),
+                       dim3(kBlock), 0, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       reinterpret_cast<__hip_bfloat16*>(y.data_ptr()),
+                       idx.data_ptr<uint8_t>(), N, C, Hi, Wi, Ho, Wo, kh,
+                       kw, sh, sw, ph, pw);
+  } else {
+    TORCH_CHECK(x.scalar_type() == at::kFloat);
+    hipLaunchKernelGGL(maxpool_fwd_kernel<float>, dim3(grid), dim3(kBlock),
+                       0, stream, x.data_ptr<float>(), y.data_ptr<float>(),
+                       idx.data_ptr<uint8_t>(), N, C, Hi, Wi, Ho, Wo, kh,
+                       kw, sh, sw, ph, pw);
+  }
+  return {y, idx};
+}
+
+at::Tensor maxpool_bwd(const at::Tensor& dy_in, const at::Tensor& idx,
+                       int N, int C, int Hi, int Wi, int kh, int kw, int sh,
+                       int sw, int ph, int pw) {
+  auto dy = dy_in.contiguous(at::MemoryFormat::ChannelsLast);
+  int Ho = dy.size(2), Wo = dy.size(3);
+  int vn = dy.scalar_type() == at::kBFloat16 ? 8 : 4;
+  auto dx = at::empty({N, C, Hi, Wi},
+                      dy.options().memory_format(
+                          at::MemoryFormat::ChannelsLast));
+  int64_t total = (int64_t)N * Hi * Wi * (C / vn);
+  int grid = elementwise_grid(total, kBlock, 2);
+  auto stream = at::hip::getCurrentHIPStream();
+  if (dy.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(maxpool_bwd_kernel<__hip_bfloat16>, dim3(grid),
+                       dim3(kBlock), 0, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(dy.data_ptr()),
+                       idx.data_ptr<uint8_t>(),
+                       reinterpret_cast<__hip_bfloat16*>(dx.data_ptr()), N,
+                       C, Hi, Wi, Ho, Wo, kh, kw, sh, sw, ph, pw);
+  } else {
+    hipLaunchKernelGGL(maxpool_bwd_kernel<float>, dim3(grid), dim3(kBlock),
+                       0, stream, dy.data_ptr<float>(),
+                       idx.data_ptr<uint8_t>(), dx.data_ptr<float>(), N, C,
+                       Hi, Wi, Ho, Wo, kh, kw, sh, sw, ph, pw);
+  }
+  return dx;
+}
+
+}  // namespace turboprune

@@ -24,6 +24,7 @@ import torch
 import torch.nn as nn
 
 from turboprune_amd.ops.bn import FusedBatchNorm2d, bn_act
+from turboprune_amd.ops.pool import FusedMaxPool2d
 from turboprune_amd.ops.mask_layers import Conv1dMask, ConvMask
 
 
@@ -106,7 +107,7 @@ class ResNet(nn.Module):
             self.conv1 = ConvMask(in_channels=3, out_channels=64,
                                   kernel_size=7, stride=2, padding=3,
                                   bias=False)
-            self.maxpool = nn.MaxPool2d(kernel_size=3, stride=2, padding=1)
+            self.maxpool = FusedMaxPool2d(kernel_size=3, stride=2, padding=1)
         self.bn1 = FusedBatchNorm2d(64)
         self.relu = nn.ReLU(inplace=True)
         self.layer1 = self._make_layer(block, 64, layers[0])

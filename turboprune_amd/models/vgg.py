@@ -14,6 +14,7 @@ import torch
 import torch.nn as nn
 
 from turboprune_amd.ops.bn import FusedBatchNorm2d
+from turboprune_amd.ops.pool import FusedMaxPool2d
 from turboprune_amd.ops.mask_layers import Conv1dMask, ConvMask
 
 _CFGS = {
@@ -30,7 +31,7 @@ def _make_features(cfg: List[Union[int, str]], batch_norm: bool) -> nn.Sequentia
     in_ch = 3
     for v in cfg:
         if v == "M":
-            layers.append(nn.MaxPool2d(kernel_size=2, stride=2))
+            layers.append(FusedMaxPool2d(kernel_size=2, stride=2))
         else:
             layers.append(ConvMask(in_channels=in_ch, out_channels=int(v),
                                    kernel_size=3, padding=1, bias=True))
