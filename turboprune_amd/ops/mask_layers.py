@@ -90,13 +90,37 @@ def _bias_like(bias: Optional[torch.Tensor], w: torch.Tensor):
 
 class ConvMask(_MaskedMixin, nn.Conv2d):
     """Conv2d with a multiplicative 0/1 weight mask (reference:
-    utils/mask_layers.py:10-43)."""
+    utils/mask_layers.py:10-43).
+
+    1x1 stride-1 convolutions on channels_last GPU tensors are plain
+    GEMMs (the NHWC reshape is free) and route through the in-house MFMA
+    masked GEMM when ``TURBOPRUNE_CONV1X1=gemm``; other shapes go through
+    the library conv on the cached masked weight."""
 
     def __init__(self, **kwargs) -> None:
         super().__init__(**kwargs)
         self._init_mask()
 
+    def _gemm_1x1_ok(self, x: torch.Tensor) -> bool:
+        import os
+        return (os.environ.get("TURBOPRUNE_CONV1X1", "") == "gemm"
+                and x.is_cuda and self.kernel_size == (1, 1)
+                and self.stride == (1, 1) and self.padding == (0, 0)
+                and self.groups == 1
+                and x.is_contiguous(memory_format=torch.channels_last))
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self._gemm_1x1_ok(x):
+            n, c, h, w_ = x.shape
+            x2 = x.permute(0, 2, 3, 1).reshape(-1, c)  # free view on NHWC
+            cache = self.weight_masked
+            y2 = TF.masked_linear(
+                x2, self.weight.reshape(self.out_channels, c),
+                self.mask.reshape(self.out_channels, c), self.bias,
+                cache.reshape(self.out_channels, c)
+                if cache is not None else None,
+                self.compute_dtype)
+            return y2.view(n, h, w_, self.out_channels).permute(0, 3, 1, 2)
         w = TF.masked_weight(self.weight, self.mask, self.weight_masked,
                              self.compute_dtype)
         return torch.nn.functional.conv2d(
