@@ -28,6 +28,7 @@ sources = [
     os.path.join(CSRC, "batchnorm.hip"),
     os.path.join(CSRC, "transpose.hip"),
     os.path.join(CSRC, "maxpool.hip"),
+    os.path.join(CSRC, "cifar_aug.hip"),
 ]
 
 setup(

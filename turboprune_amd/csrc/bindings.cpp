@@ -27,6 +27,8 @@ std::tuple<at::Tensor, at::Tensor> linear_bwd(const at::Tensor&,
 at::Tensor gemm_bf16(const at::Tensor&, const at::Tensor&, bool, bool);
 // fused NHWC batchnorm (batchnorm.hip)
 at::Tensor transpose2d(const at::Tensor&);
+at::Tensor crop_translate(const at::Tensor&, int64_t, const at::Tensor&);
+void cutout_(at::Tensor, const at::Tensor&, int64_t);
 std::tuple<at::Tensor, at::Tensor> maxpool_fwd(const at::Tensor&, int, int,
                                                int, int, int, int);
 at::Tensor maxpool_bwd(const at::Tensor&, const at::Tensor&, int, int, int,
@@ -66,6 +68,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA GEMM grads (grad_x, grad_w)");
   m.def("gemm_bf16", &turboprune::gemm_bf16,
         "raw MFMA bf16 GEMM (testing entry)");
+  m.def("crop_translate", &turboprune::crop_translate,
+        "CIFAR random-translate crop from reflect-padded batch");
+  m.def("cutout_", &turboprune::cutout_, "CIFAR cutout fill in-place");
   m.def("maxpool_fwd", &turboprune::maxpool_fwd,
         "NHWC maxpool forward -> (y, argmax bytes)");
   m.def("maxpool_bwd", &turboprune::maxpool_bwd,

@@ -54,6 +54,9 @@ def batch_crop_translate(padded: torch.Tensor, crop_size: int,
 
     padded: [N, C, H+2p, W+2p]; shifts: int64 [N, 2] in [0, 2p]; returns
     [N, C, crop, crop] (reference: utils/dataset.py:43-69)."""
+    if _backend.use_native(padded) and padded.dtype == torch.float32:
+        return _backend.extension().crop_translate(
+            padded.contiguous(), crop_size, shifts)
     n, c, hp, wp = padded.shape
     out = torch.empty(n, c, crop_size, crop_size, dtype=padded.dtype,
                       device=padded.device)
@@ -91,6 +94,10 @@ def batch_cutout(images: torch.Tensor, size: int,
         torch.randint(0, h, (n,), device=images.device, generator=generator),
         torch.randint(0, w, (n,), device=images.device, generator=generator),
     ], dim=1)
+    if _backend.use_native(images) and images.dtype == torch.float32 \
+            and images.is_contiguous():
+        _backend.extension().cutout_(images, centers, size)
+        return images
     mask = make_cutout_mask(n, h, w, size, centers, images.device)
     images.masked_fill_(mask.unsqueeze(1), 0.0)
     return images
