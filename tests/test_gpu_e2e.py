@@ -157,3 +157,23 @@ def test_resnet50_magnitude_prune_at_scale():
                    for _, mm in masked_modules(pm.model)
                    if int(mm.mask.sum()) > 0)
     assert kept_min > 0
+
+
+def test_patch_embed_gemm_matches_conv():
+    """PatchEmbed's GEMM path must equal the conv path exactly
+    (fp32, masked)."""
+    from turboprune_amd.models.deit import PatchEmbed
+    torch.manual_seed(0)
+    pe = PatchEmbed(img_size=64, patch_size=16, in_chans=3,
+                    embed_dim=128).to("cuda:0")
+    pe.proj.mask.bernoulli_(0.7)
+    x = torch.randn(2, 3, 64, 64, device="cuda:0", requires_grad=True)
+    y = pe(x)  # GEMM path (cuda)
+    # conv oracle
+    w = pe.proj.weight * pe.proj.mask
+    y_ref = torch.nn.functional.conv2d(
+        x.detach(), w, pe.proj.bias, stride=16).flatten(2).transpose(1, 2)
+    assert (y - y_ref).abs().max().item() < 1e-3
+    y.sum().backward()
+    assert pe.proj.weight.grad is not None
+    assert torch.all(pe.proj.weight.grad[pe.proj.mask == 0] == 0)

@@ -26,6 +26,12 @@ from turboprune_amd.ops.mask_layers import Conv1dMask, ConvMask
 
 
 class PatchEmbed(nn.Module):
+    """Patchify + project. The stride-P PxP conv is mathematically a GEMM
+    over non-overlapping patches; on GPU it runs on the masked MFMA GEMM
+    (MIOpen's only kernel for this 3-channel odd shape is a ~60 ms naive
+    fallback on some boxes). The parameter stays a ConvMask for
+    state-dict compatibility."""
+
     def __init__(self, img_size=224, patch_size=16, in_chans=3, embed_dim=768):
         super().__init__()
         self.img_size = (img_size, img_size)
@@ -36,6 +42,24 @@ class PatchEmbed(nn.Module):
                              bias=True)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        B, C, H, W = x.shape
+        P = self.patch_size[0]
+        if x.is_cuda and H % P == 0 and W % P == 0:
+            from turboprune_amd.ops import functional as TF
+            E = self.proj.out_channels
+            # patches ordered (row-in-patch, col-in-patch, channel) to
+            # match a channels_last-reshaped weight
+            xp = (x.permute(0, 2, 3, 1)
+                   .reshape(B, H // P, P, W // P, P, C)
+                   .permute(0, 1, 3, 2, 4, 5)
+                   .reshape(B, (H // P) * (W // P), P * P * C))
+            w2 = self.proj.weight.permute(0, 2, 3, 1).reshape(E, -1)
+            m2 = self.proj.mask.permute(0, 2, 3, 1).reshape(E, -1)
+            cache = self.proj.weight_masked
+            c2 = cache.permute(0, 2, 3, 1).reshape(E, -1) \
+                if cache is not None else None
+            return TF.masked_linear(xp, w2, m2, self.proj.bias, c2,
+                                    self.proj.compute_dtype)
         x = self.proj(x)                      # B, C, H/ps, W/ps
         return x.flatten(2).transpose(1, 2)   # B, N, C
 
