@@ -29,6 +29,7 @@ sources = [
     os.path.join(CSRC, "transpose.hip"),
     os.path.join(CSRC, "maxpool.hip"),
     os.path.join(CSRC, "cifar_aug.hip"),
+    os.path.join(CSRC, "layernorm_gelu.hip"),
 ]
 
 setup(

@@ -101,3 +101,56 @@ def test_world_info_defaults():
     from turboprune_amd.parallel.ddp import world_info
     rank, local, world = world_info()
     assert (rank, local, world) == (0, 0, 1)
+
+
+def _worker_train_parity(rank, world, port, q):
+    """Full masked train step under DDP (gloo): 2 ranks each with half
+    the batch must equal 1 process with the full batch."""
+    _init(rank, world, port)
+    from turboprune_amd.parallel.ddp import wrap_ddp
+    from turboprune_amd.optim import FusedMaskedSGD
+
+    model = _tiny_model(seed=7)
+    with torch.no_grad():
+        model[0].mask.bernoulli_(0.5)
+        model[2].mask.bernoulli_(0.5)
+    ddp = wrap_ddp(model)
+    opt = FusedMaskedSGD(model.parameters(), lr=0.1, momentum=0.9,
+                         weight_decay=1e-3, model=model)
+    torch.manual_seed(99)
+    for step in range(3):
+        x = torch.randn(4 * world, 8)
+        y = torch.randn(4 * world, 4)
+        xi = x[rank * 4:(rank + 1) * 4]
+        yi = y[rank * 4:(rank + 1) * 4]
+        loss = torch.nn.functional.mse_loss(ddp(xi), yi)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+    q.put(("w", rank, model[0].weight.detach().tolist()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_ddp_masked_train_step_matches_single_process():
+    results = _run_workers(_worker_train_parity)
+    weights = {rank: torch.tensor(w) for (_, rank, w) in results}
+    assert torch.allclose(weights[0], weights[1], atol=1e-6)
+
+    # single-process reference (rank-0 init, full batch each step)
+    from turboprune_amd.optim import FusedMaskedSGD
+    model = _tiny_model(seed=7)
+    with torch.no_grad():
+        model[0].mask.bernoulli_(0.5)
+        model[2].mask.bernoulli_(0.5)
+    opt = FusedMaskedSGD(model.parameters(), lr=0.1, momentum=0.9,
+                         weight_decay=1e-3, model=model)
+    torch.manual_seed(99)
+    for step in range(3):
+        x = torch.randn(8, 8)
+        y = torch.randn(8, 4)
+        loss = torch.nn.functional.mse_loss(model(x), y)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+    assert torch.allclose(weights[0], model[0].weight.detach(), atol=1e-5)

@@ -306,3 +306,47 @@ def test_cutout_kernel(ext):
     mask = make_cutout_mask(4, 32, 32, 8, centers, torch.device(DEV))
     ref.masked_fill_(mask.unsqueeze(1), 0.0)
     assert torch.equal(imgs, ref)
+
+
+# ---------------------------------------------------------------- LN + GELU
+@pytest.mark.parametrize("C", [384, 768])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_layernorm_matches_torch(ext, C, dtype):
+    from turboprune_amd.ops.norm_act import FusedLayerNorm
+    torch.manual_seed(0)
+    ln = FusedLayerNorm(C, eps=1e-6).to(DEV)
+    ref = torch.nn.LayerNorm(C, eps=1e-6).to(DEV)
+    ref.load_state_dict(ln.state_dict())
+    x = torch.randn(4, 50, C, device=DEV).to(dtype).requires_grad_()
+    x2 = x.detach().clone().requires_grad_()
+    y = ln(x)
+    y_ref = ref(x2.float())
+    tol = 1e-5 if dtype == torch.float32 else 3e-2
+    assert (y.float() - y_ref).abs().max().item() < tol
+    dy = torch.randn_like(y_ref).to(dtype)
+    y.backward(dy)
+    y_ref.backward(dy.float())
+    gtol = 1e-4 if dtype == torch.float32 else 6e-2
+    assert (x.grad.float() - x2.grad).abs().max().item() < gtol
+    assert torch.allclose(ln.weight.grad, ref.weight.grad, atol=1e-2,
+                          rtol=1e-2)
+    assert torch.allclose(ln.bias.grad, ref.bias.grad, atol=1e-2,
+                          rtol=1e-2)
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_fused_gelu_matches_torch(ext, dtype):
+    from turboprune_amd.ops.norm_act import FusedGELU
+    torch.manual_seed(1)
+    g = FusedGELU()
+    x = torch.randn(64, 1536, device=DEV).to(dtype).requires_grad_()
+    x2 = x.detach().clone().requires_grad_()
+    y = g(x)
+    y_ref = torch.nn.functional.gelu(x2.float())
+    tol = 1e-6 if dtype == torch.float32 else 2e-2
+    assert (y.float() - y_ref).abs().max().item() < tol
+    dy = torch.randn_like(y_ref).to(dtype)
+    y.backward(dy)
+    y_ref.backward(dy.float())
+    assert (x.grad.float() - x2.grad).abs().max().item() < \
+        (1e-4 if dtype == torch.float32 else 3e-2)

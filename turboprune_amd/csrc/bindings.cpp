@@ -27,6 +27,13 @@ std::tuple<at::Tensor, at::Tensor> linear_bwd(const at::Tensor&,
 at::Tensor gemm_bf16(const at::Tensor&, const at::Tensor&, bool, bool);
 // fused NHWC batchnorm (batchnorm.hip)
 at::Tensor transpose2d(const at::Tensor&);
+std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_fwd(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&, double);
+std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_bwd(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&);
+at::Tensor gelu_fwd(const at::Tensor&);
+at::Tensor gelu_bwd(const at::Tensor&, const at::Tensor&);
 at::Tensor crop_translate(const at::Tensor&, int64_t, const at::Tensor&);
 void cutout_(at::Tensor, const at::Tensor&, int64_t);
 std::tuple<at::Tensor, at::Tensor> maxpool_fwd(const at::Tensor&, int, int,
@@ -68,6 +75,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA GEMM grads (grad_x, grad_w)");
   m.def("gemm_bf16", &turboprune::gemm_bf16,
         "raw MFMA bf16 GEMM (testing entry)");
+  m.def("ln_fwd", &turboprune::ln_fwd, "fused LayerNorm fwd");
+  m.def("ln_bwd", &turboprune::ln_bwd, "fused LayerNorm bwd");
+  m.def("gelu_fwd", &turboprune::gelu_fwd, "fused exact GELU fwd");
+  m.def("gelu_bwd", &turboprune::gelu_bwd, "fused exact GELU bwd");
   m.def("crop_translate", &turboprune::crop_translate,
         "CIFAR random-translate crop from reflect-padded batch");
   m.def("cutout_", &turboprune::cutout_, "CIFAR cutout fill in-place");

@@ -23,6 +23,7 @@ import torch
 import torch.nn as nn
 
 from turboprune_amd.ops.mask_layers import Conv1dMask, ConvMask
+from turboprune_amd.ops.norm_act import FusedGELU, FusedLayerNorm
 
 
 class PatchEmbed(nn.Module):
@@ -91,7 +92,7 @@ class Mlp(nn.Module):
     def __init__(self, in_features, hidden_features, drop=0.0):
         super().__init__()
         self.fc1 = Conv1dMask(in_features, hidden_features, bias=True)
-        self.act = nn.GELU()
+        self.act = FusedGELU()
         self.fc2 = Conv1dMask(hidden_features, in_features, bias=True)
         self.drop = nn.Dropout(drop)
 
@@ -120,7 +121,7 @@ class VisionTransformer(nn.Module):
                  mlp_ratio=4.0, qkv_bias=True, drop_rate=0.0,
                  attn_drop_rate=0.0, norm_layer=None):
         super().__init__()
-        norm_layer = norm_layer or partial(nn.LayerNorm, eps=1e-6)
+        norm_layer = norm_layer or partial(FusedLayerNorm, eps=1e-6)
         self.num_classes = num_classes
         self.embed_dim = embed_dim
         self.patch_embed = PatchEmbed(img_size, patch_size, in_chans, embed_dim)
@@ -207,7 +208,7 @@ def _deit(embed_dim, depth, num_heads, img_size=224, distilled=False,
     return cls(img_size=img_size, patch_size=16, embed_dim=embed_dim,
                depth=depth, num_heads=num_heads, mlp_ratio=4.0,
                qkv_bias=True, num_classes=num_classes,
-               norm_layer=partial(nn.LayerNorm, eps=1e-6), **kwargs)
+               norm_layer=partial(FusedLayerNorm, eps=1e-6), **kwargs)
 
 
 # factory names mirror the reference's timm registrations (utils/deit.py:69-253)
