@@ -42,7 +42,10 @@ TP_DEVICE void ln_store(T* p, const float* v) {
 }
 
 // ---------------- LayerNorm forward --------------------------------------
-// block = 256 (4 waves); wave w handles rows blockIdx.x*4 + w, grid-stride
+// block = 256 (4 waves); wave w handles rows blockIdx.x*4 + w, grid-stride.
+// Each lane caches its <=MAXPOS octets in registers: x is read ONCE.
+constexpr int kLnMaxPos = 4;  // supports C <= 64*kLnMaxPos*VN
+
 template <typename T>
 __global__ void ln_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                               const float* __restrict__ gamma,
@@ -57,12 +60,19 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
   for (int64_t row = (int64_t)blockIdx.x * 4 + wid; row < rows;
        row += (int64_t)gridDim.x * 4) {
     const T* xr = x + row * C;
+    float v[kLnMaxPos][VN];
     float s = 0.f, ss = 0.f;
-    for (int o = lane; o < n_oct; o += kWave) {
-      float v[VN];
-      ln_load(xr + o * VN, v);
 #pragma unroll
-      for (int j = 0; j < VN; ++j) { s += v[j]; ss += v[j] * v[j]; }
+    for (int p = 0; p < kLnMaxPos; ++p) {
+      int o = lane + p * kWave;
+      if (o < n_oct) {
+        ln_load(xr + o * VN, v[p]);
+#pragma unroll
+        for (int j = 0; j < VN; ++j) {
+          s += v[p][j];
+          ss += v[p][j] * v[p][j];
+        }
+      }
     }
     s = wave_reduce_sum(s);
     ss = wave_reduce_sum(ss);
@@ -73,22 +83,27 @@ __global__ void ln_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
     float rstd = rsqrtf(var + eps);
     if (lane == 0) { mean_out[row] = mean; rstd_out[row] = rstd; }
     T* yr = y + row * C;
-    for (int o = lane; o < n_oct; o += kWave) {
-      float v[VN], out[VN];
-      ln_load(xr + o * VN, v);
 #pragma unroll
-      for (int j = 0; j < VN; ++j) {
-        int c = o * VN + j;
-        out[j] = (v[j] - mean) * rstd * gamma[c] + beta[c];
+    for (int p = 0; p < kLnMaxPos; ++p) {
+      int o = lane + p * kWave;
+      if (o < n_oct) {
+        float out[VN];
+#pragma unroll
+        for (int j = 0; j < VN; ++j) {
+          int c = o * VN + j;
+          out[j] = (v[p][j] - mean) * rstd * gamma[c] + beta[c];
+        }
+        ln_store(yr + o * VN, out);
       }
-      ln_store(yr + o * VN, out);
     }
   }
 }
 
 // ---------------- LayerNorm backward --------------------------------------
 // dx = rstd * gamma .* dy - rstd/C * (sum(dy.*gamma) + xhat*sum(dy.*gamma.*xhat))
-// per-channel dgamma/dbeta partials: LDS [C] accumulators per block
+// Each lane owns fixed channel octets (o = lane + p*64), so dgamma/dbeta
+// accumulate in REGISTERS across the wave's rows — no atomics; each WAVE
+// writes its own partial row, summed by the finalize kernel.
 template <typename T>
 __global__ void ln_bwd_kernel(const T* __restrict__ x,
                               const T* __restrict__ dy, T* __restrict__ dx,
@@ -99,38 +114,38 @@ __global__ void ln_bwd_kernel(const T* __restrict__ x,
                               float* __restrict__ partial_db, int64_t rows,
                               int C) {
   constexpr int VN = LnVec<T>::kN;
-  extern __shared__ __attribute__((aligned(16))) float lds[];
-  float* ldg = lds;          // [C]
-  float* ldb = lds + C;      // [C]
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    ldg[c] = 0.f;
-    ldb[c] = 0.f;
-  }
-  __syncthreads();
-
   int lane = threadIdx.x & (kWave - 1);
   int wid = threadIdx.x / kWave;
   int n_oct = C / VN;
+  float adg[kLnMaxPos][VN], adb[kLnMaxPos][VN];
+#pragma unroll
+  for (int p = 0; p < kLnMaxPos; ++p)
+#pragma unroll
+    for (int j = 0; j < VN; ++j) adg[p][j] = adb[p][j] = 0.f;
+
   for (int64_t row = (int64_t)blockIdx.x * 4 + wid; row < rows;
        row += (int64_t)gridDim.x * 4) {
     const T* xr = x + row * C;
     const T* gr = dy + row * C;
     float mean = mean_in[row], rstd = rstd_in[row];
+    float v[kLnMaxPos][VN], g[kLnMaxPos][VN];
     float s1 = 0.f, s2 = 0.f;  // sum(dy*gamma), sum(dy*gamma*xhat)
-    for (int o = lane; o < n_oct; o += kWave) {
-      float v[VN], g[VN];
-      ln_load(xr + o * VN, v);
-      ln_load(gr + o * VN, g);
 #pragma unroll
-      for (int j = 0; j < VN; ++j) {
-        int c = o * VN + j;
-        float xhat = (v[j] - mean) * rstd;
-        float dg = g[j] * gamma[c];
-        s1 += dg;
-        s2 += dg * xhat;
-        // per-channel partials into LDS (atomic: 4 waves share)
-        atomicAdd(&ldg[c], g[j] * xhat);
-        atomicAdd(&ldb[c], g[j]);
+    for (int p = 0; p < kLnMaxPos; ++p) {
+      int o = lane + p * kWave;
+      if (o < n_oct) {
+        ln_load(xr + o * VN, v[p]);
+        ln_load(gr + o * VN, g[p]);
+#pragma unroll
+        for (int j = 0; j < VN; ++j) {
+          int c = o * VN + j;
+          float xhat = (v[p][j] - mean) * rstd;
+          float dgj = g[p][j] * gamma[c];
+          s1 += dgj;
+          s2 += dgj * xhat;
+          adg[p][j] += g[p][j] * xhat;
+          adb[p][j] += g[p][j];
+        }
       }
     }
     s1 = wave_reduce_sum(s1);
@@ -138,23 +153,33 @@ __global__ void ln_bwd_kernel(const T* __restrict__ x,
     s1 = __shfl(s1, 0, kWave) / C;
     s2 = __shfl(s2, 0, kWave) / C;
     T* dxr = dx + row * C;
-    for (int o = lane; o < n_oct; o += kWave) {
-      float v[VN], g[VN], out[VN];
-      ln_load(xr + o * VN, v);
-      ln_load(gr + o * VN, g);
 #pragma unroll
-      for (int j = 0; j < VN; ++j) {
-        int c = o * VN + j;
-        float xhat = (v[j] - mean) * rstd;
-        out[j] = rstd * (g[j] * gamma[c] - s1 - xhat * s2);
+    for (int p = 0; p < kLnMaxPos; ++p) {
+      int o = lane + p * kWave;
+      if (o < n_oct) {
+        float out[VN];
+#pragma unroll
+        for (int j = 0; j < VN; ++j) {
+          int c = o * VN + j;
+          float xhat = (v[p][j] - mean) * rstd;
+          out[j] = rstd * (g[p][j] * gamma[c] - s1 - xhat * s2);
+        }
+        ln_store(dxr + o * VN, out);
       }
-      ln_store(dxr + o * VN, out);
     }
   }
-  __syncthreads();
-  for (int c = threadIdx.x; c < C; c += blockDim.x) {
-    partial_dg[(int64_t)blockIdx.x * C + c] = ldg[c];
-    partial_db[(int64_t)blockIdx.x * C + c] = ldb[c];
+  // per-wave partial rows: slot = blockIdx.x*4 + wid
+  int64_t slot = (int64_t)blockIdx.x * 4 + wid;
+#pragma unroll
+  for (int p = 0; p < kLnMaxPos; ++p) {
+    int o = lane + p * kWave;
+    if (o < n_oct) {
+#pragma unroll
+      for (int j = 0; j < VN; ++j) {
+        partial_dg[slot * C + o * VN + j] = adg[p][j];
+        partial_db[slot * C + o * VN + j] = adb[p][j];
+      }
+    }
   }
 }
 
@@ -188,6 +213,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_fwd(
   int64_t rows = x.numel() / C;
   int vn = x.scalar_type() == at::kBFloat16 ? 8 : 4;
   TORCH_CHECK(C % vn == 0, "ln_fwd: C % ", vn, " != 0");
+  TORCH_CHECK(C / vn <= 64 * 4, "ln_fwd: C too large for register cache");
   auto fopt = x.options().dtype(at::kFloat);
   auto y = at::empty_like(x);
   auto mean = at::empty({rows}, fopt);
@@ -227,10 +253,11 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_bwd(
   auto dx = at::empty_like(x);
   auto g = gamma.contiguous().to(at::kFloat);
   int grid = ln_grid(rows);
-  auto partial_dg = at::empty({grid, C}, fopt);
-  auto partial_db = at::empty({grid, C}, fopt);
+  int slots = grid * 4;  // one partial row per wave
+  auto partial_dg = at::empty({slots, C}, fopt);
+  auto partial_db = at::empty({slots, C}, fopt);
   auto stream = at::hip::getCurrentHIPStream();
-  size_t lds_bytes = 2 * C * sizeof(float);
+  size_t lds_bytes = 0;
   if (x.scalar_type() == at::kBFloat16) {
     hipLaunchKernelGGL(ln_bwd_kernel<__hip_bfloat16>, dim3(grid),
                        dim3(kBlock), lds_bytes, stream,
@@ -254,7 +281,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_bwd(
   hipLaunchKernelGGL(ln_param_reduce_kernel, dim3(cblocks), dim3(kBlock), 0,
                      stream, partial_dg.data_ptr<float>(),
                      partial_db.data_ptr<float>(), dgamma.data_ptr<float>(),
-                     dbeta.data_ptr<float>(), C, grid);
+                     dbeta.data_ptr<float>(), C, slots);
   return {dx, dgamma, dbeta};
 }
 
