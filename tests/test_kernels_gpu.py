@@ -350,3 +350,26 @@ def test_fused_gelu_matches_torch(ext, dtype):
     y_ref.backward(dy.float())
     assert (x.grad.float() - x2.grad).abs().max().item() < \
         (1e-4 if dtype == torch.float32 else 3e-2)
+
+
+def test_fused_layernorm_bench_shape(ext):
+    """LN at the DeiT bench shape (50432 rows) incl. backward — guards
+    the grid-stride/partial indexing at scale."""
+    from turboprune_amd.ops.norm_act import FusedLayerNorm
+    torch.manual_seed(0)
+    ln = FusedLayerNorm(384, eps=1e-6).to(DEV)
+    x = torch.randn(256, 197, 384, device=DEV).to(torch.bfloat16) \
+        .requires_grad_()
+    y = ln(x)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    torch.cuda.synchronize()
+    ref = torch.nn.LayerNorm(384, eps=1e-6).to(DEV)
+    ref.load_state_dict(ln.state_dict())
+    x2 = x.detach().float().requires_grad_()
+    y_ref = ref(x2)
+    y_ref.backward(dy.float())
+    assert (y.float() - y_ref).abs().max().item() < 3e-2
+    assert (x.grad.float() - x2.grad).abs().max().item() < 6e-2
+    assert torch.allclose(ln.weight.grad, ref.weight.grad, rtol=2e-2,
+                          atol=2e-2)

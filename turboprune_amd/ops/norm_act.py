@@ -8,6 +8,8 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
+import os
+
 from turboprune_amd.ops import _backend
 
 
@@ -33,7 +35,8 @@ class FusedLayerNorm(nn.LayerNorm):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         vn = 8 if x.dtype == torch.bfloat16 else 4
-        if (x.is_cuda and len(self.normalized_shape) == 1
+        if (os.environ.get("TURBOPRUNE_LN", "native") == "native"
+                and x.is_cuda and len(self.normalized_shape) == 1
                 and x.dtype in (torch.bfloat16, torch.float32)
                 and x.shape[-1] == self.normalized_shape[0]
                 and x.shape[-1] % vn == 0
@@ -62,7 +65,8 @@ class FusedGELU(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         vn = 8 if x.dtype == torch.bfloat16 else 4
-        if (x.is_cuda and x.dtype in (torch.bfloat16, torch.float32)
+        if (os.environ.get("TURBOPRUNE_GELU", "native") == "native"
+                and x.is_cuda and x.dtype in (torch.bfloat16, torch.float32)
                 and x.numel() % vn == 0 and _backend.use_native(x)):
             return _GeluFn.apply(x)
         return F.gelu(x)
