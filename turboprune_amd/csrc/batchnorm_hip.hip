@@ -33,6 +33,8 @@
 
 namespace turboprune {
 
+void colsum_atomic(const at::Tensor&, at::Tensor);  // elementwise.hip
+
 // element vector: 16 bytes of activations
 template <typename T>
 struct Octet;
@@ -190,9 +192,8 @@ __global__ void bn_reduce_kernel(const T* __restrict__ x, int64_t rows,
 }
 
 // ---------------- forward finalize (one small launch) --------------------
-__global__ void bn_finalize_kernel(const float* __restrict__ partial_sum,
-                                   const float* __restrict__ partial_ss,
-                                   int rb,
+__global__ void bn_finalize_kernel(const float* __restrict__ sum_in,
+                                   const float* __restrict__ ss_in,
                                    const float* __restrict__ gamma,
                                    const float* __restrict__ beta,
                                    float* __restrict__ running_mean,
@@ -206,27 +207,8 @@ __global__ void bn_finalize_kernel(const float* __restrict__ partial_sum,
                                    bool update_running) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float s0 = 0.f, s1 = 0.f, s2 = 0.f, s3 = 0.f;
-  float q0 = 0.f, q1 = 0.f, q2 = 0.f, q3 = 0.f;
-  int r = 0;
-  for (; r + 3 < rb; r += 4) {
-    s0 += partial_sum[(int64_t)r * C + c];
-    s1 += partial_sum[(int64_t)(r + 1) * C + c];
-    s2 += partial_sum[(int64_t)(r + 2) * C + c];
-    s3 += partial_sum[(int64_t)(r + 3) * C + c];
-    q0 += partial_ss[(int64_t)r * C + c];
-    q1 += partial_ss[(int64_t)(r + 1) * C + c];
-    q2 += partial_ss[(int64_t)(r + 2) * C + c];
-    q3 += partial_ss[(int64_t)(r + 3) * C + c];
-  }
-  for (; r < rb; ++r) {
-    s0 += partial_sum[(int64_t)r * C + c];
-    q0 += partial_ss[(int64_t)r * C + c];
-  }
-  float sum_c = (s0 + s1) + (s2 + s3);
-  float sumsq_c = (q0 + q1) + (q2 + q3);
-  float mean = sum_c * inv_n;
-  float var = fmaxf(sumsq_c * inv_n - mean * mean, 0.f);
+  float mean = sum_in[c] * inv_n;
+  float var = fmaxf(ss_in[c] * inv_n - mean * mean, 0.f);
   float rstd = rsqrtf(var + eps);
   save_mean[c] = mean;
   save_rstd[c] = rstd;
@@ -407,9 +389,8 @@ __global__ void bn_bwd_reduce_kernel(const T* __restrict__ x,
   }
 }
 
-__global__ void bn_bwd_finalize_kernel(const float* __restrict__ partial_g,
-                                       const float* __restrict__ partial_gx,
-                                       int rb,
+__global__ void bn_bwd_finalize_kernel(const float* __restrict__ sum_g_in,
+                                       const float* __restrict__ sum_gx_in,
                                        const float* __restrict__ gamma,
                                        const float* __restrict__ save_mean,
                                        const float* __restrict__ save_rstd,
@@ -421,26 +402,8 @@ __global__ void bn_bwd_finalize_kernel(const float* __restrict__ partial_g,
                                        float inv_n) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float g0 = 0.f, g1 = 0.f, g2 = 0.f, g3 = 0.f;
-  float x0 = 0.f, x1 = 0.f, x2 = 0.f, x3 = 0.f;
-  int r = 0;
-  for (; r + 3 < rb; r += 4) {
-    g0 += partial_g[(int64_t)r * C + c];
-    g1 += partial_g[(int64_t)(r + 1) * C + c];
-    g2 += partial_g[(int64_t)(r + 2) * C + c];
-    g3 += partial_g[(int64_t)(r + 3) * C + c];
-    x0 += partial_gx[(int64_t)r * C + c];
-    x1 += partial_gx[(int64_t)(r + 1) * C + c];
-    x2 += partial_gx[(int64_t)(r + 2) * C + c];
-    x3 += partial_gx[(int64_t)(r + 3) * C + c];
-  }
-  for (; r < rb; ++r) {
-    g0 += partial_g[(int64_t)r * C + c];
-    x0 += partial_gx[(int64_t)r * C + c];
-  }
-  float sum_g_c = (g0 + g1) + (g2 + g3);
-  float sum_gx_c = (x0 + x1) + (x2 + x3);
-  float sum_dy = sum_g_c;
+  float sum_dy = sum_g_in[c];
+  float sum_gx_c = sum_gx_in[c];
   // Σ g·xhat = rstd · (Σ g·x − mean·Σ g)
   float sum_dy_xhat = save_rstd[c] * (sum_gx_c - save_mean[c] * sum_dy);
   dgamma[c] = sum_dy_xhat;
@@ -589,9 +552,13 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_fwd(
 #undef BN_RED
     float inv_n = 1.0f / (float)sh.rows;
     float unbiased = sh.rows > 1 ? (float)sh.rows / (sh.rows - 1) : 1.0f;
+    auto sum = at::zeros({sh.C}, fopt);
+    auto sumsq = at::zeros({sh.C}, fopt);
+    colsum_atomic(partial_sum, sum);
+    colsum_atomic(partial_ss, sumsq);
     hipLaunchKernelGGL(bn_finalize_kernel, dim3(cblocks), dim3(kBlock), 0,
-                       stream, partial_sum.data_ptr<float>(),
-                       partial_ss.data_ptr<float>(), (int)rg.x,
+                       stream, sum.data_ptr<float>(),
+                       sumsq.data_ptr<float>(),
                        g.data_ptr<float>(),
                        b.data_ptr<float>(),
                        running_mean.data_ptr<float>(),
@@ -696,9 +663,13 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd(
   auto c_xhat = at::empty({sh.C}, fopt);
   auto c_const = at::empty({sh.C}, fopt);
   int cblocks = (sh.C + kBlock - 1) / kBlock;
+  auto sg = at::zeros({sh.C}, fopt);
+  auto sgx = at::zeros({sh.C}, fopt);
+  colsum_atomic(sum_dy, sg);
+  colsum_atomic(sum_dy_xhat, sgx);
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(cblocks), dim3(kBlock), 0,
-                     stream, sum_dy.data_ptr<float>(),
-                     sum_dy_xhat.data_ptr<float>(), (int)rg.x,
+                     stream, sg.data_ptr<float>(),
+                     sgx.data_ptr<float>(),
                      g.data_ptr<float>(),
                      save_mean.data_ptr<float>(), save_rstd.data_ptr<float>(),
                      dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),

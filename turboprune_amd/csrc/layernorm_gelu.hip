@@ -14,6 +14,8 @@
 
 namespace turboprune {
 
+void colsum_atomic(const at::Tensor&, at::Tensor);  // elementwise.hip
+
 template <typename T>
 struct LnVec;
 template <>
@@ -274,13 +276,10 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_bwd(
                        rstd.data_ptr<float>(), partial_dg.data_ptr<float>(),
                        partial_db.data_ptr<float>(), rows, C);
   }
-  auto dgamma = at::empty({C}, fopt);
-  auto dbeta = at::empty({C}, fopt);
-  int cblocks = (C + kBlock - 1) / kBlock;
-  hipLaunchKernelGGL(ln_param_reduce_kernel, dim3(cblocks), dim3(kBlock), 0,
-                     stream, partial_dg.data_ptr<float>(),
-                     partial_db.data_ptr<float>(), dgamma.data_ptr<float>(),
-                     dbeta.data_ptr<float>(), C, slots);
+  auto dgamma = at::zeros({C}, fopt);
+  auto dbeta = at::zeros({C}, fopt);
+  colsum_atomic(partial_dg, dgamma);
+  colsum_atomic(partial_db, dbeta);
   return {dx, dgamma, dbeta};
 }
 
