@@ -103,3 +103,31 @@ def test_unwrap_inner_model():
     cfg = compose("cifar10_er_erk")
     pm = build_model(cfg)
     assert unwrap_inner_model(pm) is pm.model
+
+
+def test_reset_optimizer_roundtrip(tmp_path):
+    import torch.nn as nn
+
+    from turboprune_amd.utils.console import reset_optimizer
+    cfg = _cfg(tmp_path)
+    prefix, expt_dir = gen_expt_dir(cfg)
+    model = nn.Linear(4, 4)
+    opt = torch.optim.SGD(model.parameters(), lr=0.1, momentum=0.9)
+    model(torch.randn(2, 4)).sum().backward()
+    opt.step()
+    torch.save(opt.state_dict(),
+               os.path.join(expt_dir, "artifacts", "optimizer_init.pt"))
+    # perturb state then rewind
+    model(torch.randn(2, 4)).sum().backward()
+    opt.step()
+    reset_optimizer(expt_dir, opt, "imp")
+    ref = torch.load(os.path.join(expt_dir, "artifacts",
+                                  "optimizer_init.pt"), weights_only=False)
+    assert str(opt.state_dict()["state"]) == str(ref["state"])
+
+
+def test_display_training_info_smoke(capsys):
+    from turboprune_amd.utils.console import display_training_info
+    display_training_info({"device": "cuda:0", "expt_dir": "/a/b/c"},
+                          {"lr": 0.2}, cycle_info={"cycle": 1},
+                          training_info={"epochs": 10})

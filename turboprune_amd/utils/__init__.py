@@ -7,3 +7,7 @@ from turboprune_amd.utils.experiment import (  # noqa: F401
     resume_experiment,
 )
 from turboprune_amd.utils.logging import MetricsLogger, Throughput  # noqa: F401
+from turboprune_amd.utils.console import (  # noqa: F401
+    display_training_info,
+    reset_optimizer,
+)
