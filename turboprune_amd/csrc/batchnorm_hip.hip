@@ -33,7 +33,9 @@
 
 namespace turboprune {
 
-void colsum_atomic(const at::Tensor&, at::Tensor);  // elementwise.hip
+void colsum_atomic(const at::Tensor&, at::Tensor);   // elementwise.hip
+void colsum2_atomic(const at::Tensor&, const at::Tensor&, at::Tensor,
+                    at::Tensor);  // elementwise.hip
 
 // element vector: 16 bytes of activations
 template <typename T>
@@ -554,8 +556,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_fwd(
     float unbiased = sh.rows > 1 ? (float)sh.rows / (sh.rows - 1) : 1.0f;
     auto sum = at::zeros({sh.C}, fopt);
     auto sumsq = at::zeros({sh.C}, fopt);
-    colsum_atomic(partial_sum, sum);
-    colsum_atomic(partial_ss, sumsq);
+    colsum2_atomic(partial_sum, partial_ss, sum, sumsq);
     hipLaunchKernelGGL(bn_finalize_kernel, dim3(cblocks), dim3(kBlock), 0,
                        stream, sum.data_ptr<float>(),
                        sumsq.data_ptr<float>(),
@@ -665,8 +666,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd(
   int cblocks = (sh.C + kBlock - 1) / kBlock;
   auto sg = at::zeros({sh.C}, fopt);
   auto sgx = at::zeros({sh.C}, fopt);
-  colsum_atomic(sum_dy, sg);
-  colsum_atomic(sum_dy_xhat, sgx);
+  colsum2_atomic(sum_dy, sum_dy_xhat, sg, sgx);
   hipLaunchKernelGGL(bn_bwd_finalize_kernel, dim3(cblocks), dim3(kBlock), 0,
                      stream, sg.data_ptr<float>(),
                      sgx.data_ptr<float>(),

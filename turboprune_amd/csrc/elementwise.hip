@@ -263,4 +263,45 @@ void colsum_atomic(const at::Tensor& partial, at::Tensor out) {
                      out.data_ptr<float>(), R, C);
 }
 
+// fused pair variant: one launch for the (sum, sumsq) / (g, gx) /
+// (dgamma, dbeta) partial pairs the BN/LN finalizes always consume
+// together (212 single-array launches per ResNet50 step otherwise)
+__global__ void colsum2_atomic_kernel(const float* __restrict__ in_a,
+                                      const float* __restrict__ in_b,
+                                      float* __restrict__ out_a,
+                                      float* __restrict__ out_b, int64_t R,
+                                      int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  int64_t r0 = (int64_t)blockIdx.y * 128;
+  int64_t r1 = r0 + 128 < R ? r0 + 128 : R;
+  float a0 = 0.f, a1 = 0.f, b0 = 0.f, b1 = 0.f;
+  int64_t r = r0;
+  for (; r + 1 < r1; r += 2) {
+    a0 += in_a[r * C + c];
+    a1 += in_a[(r + 1) * C + c];
+    b0 += in_b[r * C + c];
+    b1 += in_b[(r + 1) * C + c];
+  }
+  for (; r < r1; ++r) {
+    a0 += in_a[r * C + c];
+    b0 += in_b[r * C + c];
+  }
+  atomicAdd(&out_a[c], a0 + a1);
+  atomicAdd(&out_b[c], b0 + b1);
+}
+
+void colsum2_atomic(const at::Tensor& pa, const at::Tensor& pb,
+                    at::Tensor oa, at::Tensor ob) {
+  int64_t R = pa.size(0);
+  int C = (int)pa.size(1);
+  int cblocks = (C + kBlock - 1) / kBlock;
+  int rchunks = (int)((R + 127) / 128);
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(colsum2_atomic_kernel, dim3(cblocks, rchunks),
+                     dim3(kBlock), 0, stream, pa.data_ptr<float>(),
+                     pb.data_ptr<float>(), oa.data_ptr<float>(),
+                     ob.data_ptr<float>(), R, C);
+}
+
 }  // namespace turboprune

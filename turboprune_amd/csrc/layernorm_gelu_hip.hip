@@ -15,7 +15,9 @@
 
 namespace turboprune {
 
-void colsum_atomic(const at::Tensor&, at::Tensor);  // elementwise.hip
+void colsum_atomic(const at::Tensor&, at::Tensor);   // elementwise.hip
+void colsum2_atomic(const at::Tensor&, const at::Tensor&, at::Tensor,
+                    at::Tensor);  // elementwise.hip
 
 template <typename T>
 struct LnVec;
@@ -279,8 +281,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_bwd(
   }
   auto dgamma = at::zeros({C}, fopt);
   auto dbeta = at::zeros({C}, fopt);
-  colsum_atomic(partial_dg, dgamma);
-  colsum_atomic(partial_db, dbeta);
+  colsum2_atomic(partial_dg, partial_db, dgamma, dbeta);
   return {dx, dgamma, dbeta};
 }
 
