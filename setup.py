@@ -30,6 +30,7 @@ sources = [
     os.path.join(CSRC, "maxpool.hip"),
     os.path.join(CSRC, "cifar_aug.hip"),
     os.path.join(CSRC, "layernorm_gelu.hip"),
+    os.path.join(CSRC, "rrc.hip"),
 ]
 
 setup(

@@ -133,3 +133,54 @@ def test_sharded_imagenet_reader(tmp_path):
     l1 = ShardedImageNet(str(tmp_path), "train", 16, torch.device("cpu"),
                          True, world_size=2, rank=1)
     assert set(l0.paths).isdisjoint(set(l1.paths))
+
+
+def test_sample_rrc_boxes():
+    torch.manual_seed(0)
+    boxes = augment.sample_rrc_boxes(64, 256, 256)
+    assert boxes.shape == (64, 4)
+    t, l, h, w = boxes.unbind(1)
+    assert (t >= 0).all() and (l >= 0).all()
+    assert ((t + h) <= 256).all() and ((l + w) <= 256).all()
+    assert (h > 0).all() and (w > 0).all()
+    # aspect ratio within [3/4, 4/3] (allow rounding slack)
+    ar = w.float() / h.float()
+    assert (ar > 0.7).all() and (ar < 1.43).all()
+
+
+def test_random_resized_crop_cpu_oracle():
+    torch.manual_seed(1)
+    imgs = torch.randint(0, 256, (3, 3, 64, 64), dtype=torch.uint8)
+    boxes = torch.tensor([[0, 0, 64, 64], [10, 10, 32, 32], [0, 0, 48, 64]])
+    mean = torch.tensor([0.485, 0.456, 0.406])
+    std = torch.tensor([0.229, 0.224, 0.225])
+    out = augment.random_resized_crop(imgs, boxes, mean, std, out_size=32)
+    assert out.shape == (3, 3, 32, 32)
+    # full-image box == plain resize
+    ref = torch.nn.functional.interpolate(
+        imgs[0:1].float(), size=(32, 32), mode="bilinear",
+        align_corners=False) / 255.0
+    ref = (ref - mean.view(1, 3, 1, 1)) / std.view(1, 3, 1, 1)
+    assert torch.allclose(out[0], ref[0], atol=1e-5)
+
+
+def test_sharded_reader_rrc_path(tmp_path):
+    """Shards stored at 64px, loader at 32px: train uses RRC, val center
+    crop."""
+    import subprocess
+    import sys
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for split in ("train", "val"):
+        subprocess.run([sys.executable, "scripts/make_shards.py",
+                        "--out", str(tmp_path), "--split", split,
+                        "--synthetic", "64", "--image-size", "64",
+                        "--shard-size", "64"], check=True, cwd=root)
+    from turboprune_amd.data import ShardedImageNet
+    tr = ShardedImageNet(str(tmp_path), "train", 16, torch.device("cpu"),
+                         True, image_size=32)
+    x, y = next(iter(tr))
+    assert x.shape == (16, 3, 32, 32)
+    va = ShardedImageNet(str(tmp_path), "val", 16, torch.device("cpu"),
+                         False, image_size=32)
+    xv, yv = next(iter(va))
+    assert xv.shape == (16, 3, 32, 32)

@@ -373,3 +373,17 @@ def test_fused_layernorm_bench_shape(ext):
     assert (x.grad.float() - x2.grad).abs().max().item() < 6e-2
     assert torch.allclose(ln.weight.grad, ref.weight.grad, rtol=2e-2,
                           atol=2e-2)
+
+
+def test_random_resized_crop_gpu_matches_oracle(ext):
+    from turboprune_amd.data import augment as A
+    torch.manual_seed(2)
+    imgs = torch.randint(0, 256, (4, 3, 96, 96), dtype=torch.uint8)
+    boxes = A.sample_rrc_boxes(4, 96, 96)
+    mean = torch.tensor([0.485, 0.456, 0.406])
+    std = torch.tensor([0.229, 0.224, 0.225])
+    flip = torch.tensor([True, False, True, False])
+    ref = A.random_resized_crop(imgs, boxes, mean, std, 64, flip)
+    got = A.random_resized_crop(imgs.to(DEV), boxes.to(DEV), mean.to(DEV),
+                                std.to(DEV), 64, flip.to(DEV))
+    assert (got.cpu() - ref).abs().max().item() < 1e-4

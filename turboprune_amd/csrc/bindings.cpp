@@ -35,6 +35,9 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_bwd(
 at::Tensor gelu_fwd(const at::Tensor&);
 at::Tensor gelu_bwd(const at::Tensor&, const at::Tensor&);
 at::Tensor crop_translate(const at::Tensor&, int64_t, const at::Tensor&);
+at::Tensor random_resized_crop(const at::Tensor&, const at::Tensor&,
+                               const at::Tensor&, const at::Tensor&,
+                               const at::Tensor&, int64_t, at::ScalarType);
 void cutout_(at::Tensor, const at::Tensor&, int64_t);
 std::tuple<at::Tensor, at::Tensor> maxpool_fwd(const at::Tensor&, int, int,
                                                int, int, int, int);
@@ -79,6 +82,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ln_bwd", &turboprune::ln_bwd, "fused LayerNorm bwd");
   m.def("gelu_fwd", &turboprune::gelu_fwd, "fused exact GELU fwd");
   m.def("gelu_bwd", &turboprune::gelu_bwd, "fused exact GELU bwd");
+  m.def("random_resized_crop", &turboprune::random_resized_crop,
+        "fused bilinear RandomResizedCrop + flip + normalize");
   m.def("crop_translate", &turboprune::crop_translate,
         "CIFAR random-translate crop from reflect-padded batch");
   m.def("cutout_", &turboprune::cutout_, "CIFAR cutout fill in-place");

@@ -101,3 +101,70 @@ def batch_cutout(images: torch.Tensor, size: int,
     mask = make_cutout_mask(n, h, w, size, centers, images.device)
     images.masked_fill_(mask.unsqueeze(1), 0.0)
     return images
+
+
+def sample_rrc_boxes(n: int, h: int, w: int,
+                     scale=(0.08, 1.0), ratio=(3 / 4, 4 / 3),
+                     generator: Optional[torch.Generator] = None,
+                     device: Optional[torch.device] = None) -> torch.Tensor:
+    """Sample RandomResizedCrop boxes (top, left, height, width) with
+    torchvision's area/ratio semantics (10 tries then center fallback)."""
+    import math
+    boxes = torch.empty(n, 4, dtype=torch.int64)
+    area = h * w
+    log_r = (math.log(ratio[0]), math.log(ratio[1]))
+    for i in range(n):
+        ok = False
+        for _ in range(10):
+            target = area * (torch.empty(1).uniform_(*scale,
+                                                     generator=generator)
+                             .item())
+            ar = math.exp(torch.empty(1).uniform_(*log_r,
+                                                  generator=generator)
+                          .item())
+            cw = int(round(math.sqrt(target * ar)))
+            ch = int(round(math.sqrt(target / ar)))
+            if 0 < cw <= w and 0 < ch <= h:
+                top = int(torch.randint(0, h - ch + 1, (1,),
+                                        generator=generator).item())
+                left = int(torch.randint(0, w - cw + 1, (1,),
+                                         generator=generator).item())
+                boxes[i] = torch.tensor([top, left, ch, cw])
+                ok = True
+                break
+        if not ok:
+            side = min(h, w)
+            boxes[i] = torch.tensor([(h - side) // 2, (w - side) // 2,
+                                     side, side])
+    return boxes.to(device) if device is not None else boxes
+
+
+def random_resized_crop(images_u8: torch.Tensor, boxes: torch.Tensor,
+                        mean: torch.Tensor, std: torch.Tensor,
+                        out_size: int = 224,
+                        flip: Optional[torch.Tensor] = None,
+                        out_dtype: torch.dtype = torch.float32
+                        ) -> torch.Tensor:
+    """Fused bilinear crop-resize + flip + normalize (the
+    FFCV train-pipeline equivalent, reference dataset.py:385-392)."""
+    if _backend.use_native(images_u8):
+        return _backend.extension().random_resized_crop(
+            images_u8, boxes.to(images_u8.device),
+            flip if flip is not None else torch.Tensor(),
+            mean.to(images_u8.device), std.to(images_u8.device),
+            out_size, out_dtype)
+    # torch oracle: per-image interpolate of the crop
+    outs = []
+    x = images_u8.float()
+    for i in range(images_u8.shape[0]):
+        t, l, ch, cw = boxes[i].tolist()
+        crop = x[i:i + 1, :, t:t + ch, l:l + cw]
+        out = torch.nn.functional.interpolate(
+            crop, size=(out_size, out_size), mode="bilinear",
+            align_corners=False)
+        outs.append(out)
+    out = torch.cat(outs, 0).div_(255.0)
+    out = (out - mean.view(1, -1, 1, 1)) / std.view(1, -1, 1, 1)
+    if flip is not None:
+        out[flip] = torch.flip(out[flip], dims=[-1])
+    return out.to(out_dtype)

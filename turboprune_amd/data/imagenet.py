@@ -95,7 +95,9 @@ class ShardedImageNet:
     def __init__(self, root: str, split: str, batch_size: int,
                  device: torch.device, train: bool = True,
                  world_size: int = 1, rank: int = 0, seed: int = 0,
-                 dtype: torch.dtype = torch.float32):
+                 dtype: torch.dtype = torch.float32,
+                 image_size: int = 224):
+        self.image_size = image_size
         self.dir = os.path.join(root, split)
         self.paths: List[str] = sorted(
             os.path.join(self.dir, f) for f in os.listdir(self.dir)
@@ -142,8 +144,27 @@ class ShardedImageNet:
                 if self.train:
                     flip = torch.rand(self.batch_size,
                                       device=self.device) < 0.5
-                x = augment.normalize_u8(raw, self._mean, self._std,
-                                         self.dtype, flip)
+                hs, ws = raw.shape[-2], raw.shape[-1]
+                if self.train and (hs > self.image_size
+                                   or ws > self.image_size):
+                    # FFCV-equivalent RandomResizedCrop train pipeline
+                    boxes = augment.sample_rrc_boxes(
+                        raw.shape[0], hs, ws, device=self.device)
+                    x = augment.random_resized_crop(
+                        raw, boxes, self._mean, self._std,
+                        self.image_size, flip, self.dtype)
+                elif not self.train and (hs > self.image_size
+                                         or ws > self.image_size):
+                    # center crop (val pipeline, ratio-style)
+                    t = (hs - self.image_size) // 2
+                    l = (ws - self.image_size) // 2
+                    raw = raw[..., t:t + self.image_size,
+                              l:l + self.image_size].contiguous()
+                    x = augment.normalize_u8(raw, self._mean, self._std,
+                                             self.dtype, None)
+                else:
+                    x = augment.normalize_u8(raw, self._mean, self._std,
+                                             self.dtype, flip)
                 yield x, labels[idx].to(self.device, non_blocking=True)
         self.epoch += 1
 
