@@ -38,17 +38,18 @@ __global__ void rrc_kernel(const uint8_t* __restrict__ in,
     int cw = (int)boxes[n * 4 + 3];
     int xo = (flip != nullptr && flip[n]) ? (Wo - 1 - x) : x;
 
-    // bilinear sample at crop-space position (align_corners=False)
+    // bilinear sample in CROP space (align_corners=False); clamp to the
+    // crop edges like F.interpolate on the extracted crop would
     float sy = (float)ch / Ho, sx = (float)cw / Wo;
-    float fy = ((float)y + 0.5f) * sy - 0.5f + top;
-    float fx = ((float)xo + 0.5f) * sx - 0.5f + left;
+    float fy = ((float)y + 0.5f) * sy - 0.5f;
+    float fx = ((float)xo + 0.5f) * sx - 0.5f;
     int y0 = (int)floorf(fy), x0 = (int)floorf(fx);
     float wy = fy - y0, wx = fx - x0;
     int y1 = y0 + 1, x1 = x0 + 1;
-    y0 = min(max(y0, 0), Hi - 1);
-    y1 = min(max(y1, 0), Hi - 1);
-    x0 = min(max(x0, 0), Wi - 1);
-    x1 = min(max(x1, 0), Wi - 1);
+    y0 = min(max(y0, 0), ch - 1) + top;
+    y1 = min(max(y1, 0), ch - 1) + top;
+    x0 = min(max(x0, 0), cw - 1) + left;
+    x1 = min(max(x1, 0), cw - 1) + left;
     const uint8_t* src = in + ((int64_t)n * C + c) * Hi * Wi;
     float v00 = src[(int64_t)y0 * Wi + x0];
     float v01 = src[(int64_t)y0 * Wi + x1];
