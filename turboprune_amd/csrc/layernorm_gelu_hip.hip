@@ -187,22 +187,6 @@ __global__ void ln_bwd_kernel(const T* __restrict__ x,
   }
 }
 
-__global__ void ln_param_reduce_kernel(const float* __restrict__ partial_dg,
-                                       const float* __restrict__ partial_db,
-                                       float* __restrict__ dgamma,
-                                       float* __restrict__ dbeta, int C,
-                                       int blocks) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
-  float g = 0.f, b = 0.f;
-  for (int r = 0; r < blocks; ++r) {
-    g += partial_dg[(int64_t)r * C + c];
-    b += partial_db[(int64_t)r * C + c];
-  }
-  dgamma[c] = g;
-  dbeta[c] = b;
-}
-
 // ---------------------------------------------------------------- host LN
 static int ln_grid(int64_t rows) {
   int64_t b = (rows + 3) / 4;
