@@ -12,7 +12,7 @@ implementations below are the CPU path and the numerics oracle.
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Optional
 
 import torch
 

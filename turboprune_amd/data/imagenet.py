@@ -22,7 +22,6 @@ modes:
 
 from __future__ import annotations
 
-import math
 import os
 from typing import Any, Iterator, List, Optional, Tuple
 

@@ -17,7 +17,7 @@ on MI355X (observable numbers unchanged):
 
 from __future__ import annotations
 
-from typing import Any, Dict, Iterable, List, Optional, Tuple
+from typing import Any, Iterable, List, Optional, Tuple
 
 import torch
 import torch.distributed as dist

@@ -9,7 +9,7 @@ and a Schedule column in the summary.
 
 from __future__ import annotations
 
-import os
+
 from typing import Any, Optional
 
 import torch

@@ -15,9 +15,7 @@ latent-broken. This is a working native implementation:
 
 from __future__ import annotations
 
-import math
 from functools import partial
-from typing import Optional
 
 import torch
 import torch.nn as nn

@@ -24,8 +24,7 @@ import torch.nn as nn
 from turboprune_amd.models import deit as deit_models
 from turboprune_amd.models import resnet as resnet_models
 from turboprune_amd.models import vgg as vgg_models
-from turboprune_amd.ops.mask_layers import (MASKED_LAYER_TYPES,
-                                            masked_modules)
+from turboprune_amd.ops.mask_layers import masked_modules
 
 _FACTORIES = {}
 for _mod in (resnet_models, vgg_models, deit_models):

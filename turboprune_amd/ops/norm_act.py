@@ -6,9 +6,9 @@ from __future__ import annotations
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
-
 import os
+
+import torch.nn.functional as F
 
 from turboprune_amd.ops import _backend
 

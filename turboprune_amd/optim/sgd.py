@@ -21,7 +21,7 @@ x-iterate swapped in for eval, z fast iterate in state.
 
 from __future__ import annotations
 
-from typing import Any, Dict, Iterable, Optional
+from typing import Any, Dict, Optional
 
 import torch
 import torch.nn as nn

@@ -18,12 +18,11 @@ Formats match the reference exactly so artifacts are drop-in compatible:
 
 from __future__ import annotations
 
-import math
 import os
 import random
 import uuid
 from datetime import datetime
-from typing import Any, List, Optional, Tuple
+from typing import Any, List, Tuple
 
 import numpy as np
 import torch
