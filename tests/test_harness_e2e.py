@@ -141,3 +141,57 @@ def test_resume_from_level(tmp_path):
     assert expt_dir2 == expt_dir
     assert os.path.exists(os.path.join(expt_dir, "checkpoints",
                                        "model_level_2.pt"))
+
+
+def test_snip_end_to_end(tmp_path):
+    """SNIP prune-at-init (data-driven scorer) through the full driver."""
+    cfg = compose("cifar10_er_snip", _base_overrides(
+        tmp_path, ["pruning_params.target_sparsity=0.8"]))
+    expt_dir = run(cfg)
+    sd = torch.load(os.path.join(expt_dir, "checkpoints",
+                                 "model_level_0.pt"), weights_only=True)
+    masks = [v for k, v in sd.items() if k.endswith("mask")]
+    total = sum(v.numel() for v in masks)
+    zeros = sum(int((v == 0).sum()) for v in masks)
+    assert zeros / total == pytest.approx(0.8, abs=0.02)
+
+
+def test_synflow_end_to_end(tmp_path):
+    cfg = compose("cifar10_er_synflow", _base_overrides(
+        tmp_path, ["pruning_params.target_sparsity=0.8"]))
+    expt_dir = run(cfg)
+    sd = torch.load(os.path.join(expt_dir, "checkpoints",
+                                 "model_level_0.pt"), weights_only=True)
+    masks = [v for k, v in sd.items() if k.endswith("mask")]
+    total = sum(v.numel() for v in masks)
+    zeros = sum(int((v == 0).sum()) for v in masks)
+    assert zeros / total == pytest.approx(0.8, abs=0.02)
+
+
+def test_schedule_free_harness(tmp_path):
+    cfg = compose("cifar10_er_erk", _base_overrides(tmp_path, [
+        "pruning_params.target_sparsity=0.5",
+        "optimizer_params.scheduler_type=ScheduleFree",
+    ]))
+    expt_dir = run(cfg)
+    summaries = [f for f in os.listdir(expt_dir) if f.endswith("_summary.csv")]
+    assert len(summaries) == 1
+
+
+def test_imagenet_synthetic_end_to_end(tmp_path):
+    """ImageNet-shaped synthetic run through the driver (world_size=1,
+    CPU): exercises the ImageNet loader + harness path."""
+    cfg = compose("bench_resnet50_imagenet", [
+        "model_params.model_name=resnet18",
+        "experiment_params.epochs_per_level=1",
+        "experiment_params.distributed=false",
+        "dataset_params.total_batch_size=8",
+        "+dataset_params.steps_per_epoch=2",
+        f"experiment_params.base_dir={tmp_path}/experiments",
+        f"dataset_params.data_root_dir={tmp_path}/data",
+        "pruning_params=pai_er_erk",
+        "pruning_params.target_sparsity=0.5",
+    ])
+    expt_dir = run(cfg)
+    assert os.path.exists(os.path.join(expt_dir, "checkpoints",
+                                       "model_level_0.pt"))
