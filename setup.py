@@ -31,6 +31,7 @@ sources = [
     os.path.join(CSRC, "cifar_aug.hip"),
     os.path.join(CSRC, "layernorm_gelu.hip"),
     os.path.join(CSRC, "rrc.hip"),
+    os.path.join(CSRC, "conv_implicit.hip"),
 ]
 
 setup(

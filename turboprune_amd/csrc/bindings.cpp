@@ -27,6 +27,9 @@ std::tuple<at::Tensor, at::Tensor> linear_bwd(const at::Tensor&,
 at::Tensor gemm_bf16(const at::Tensor&, const at::Tensor&, bool, bool);
 // fused NHWC batchnorm (batchnorm.hip)
 at::Tensor transpose2d(const at::Tensor&);
+at::Tensor conv2d_implicit_fwd(const at::Tensor&, const at::Tensor&,
+                               const c10::optional<at::Tensor>&, int64_t,
+                               int64_t);
 std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_fwd(
     const at::Tensor&, const at::Tensor&, const at::Tensor&, double);
 std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_bwd(
@@ -82,6 +85,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ln_bwd", &turboprune::ln_bwd, "fused LayerNorm bwd");
   m.def("gelu_fwd", &turboprune::gelu_fwd, "fused exact GELU fwd");
   m.def("gelu_bwd", &turboprune::gelu_bwd, "fused exact GELU bwd");
+  m.def("conv2d_implicit_fwd", &turboprune::conv2d_implicit_fwd,
+        "implicit-GEMM NHWC conv forward (experimental)");
   m.def("random_resized_crop", &turboprune::random_resized_crop,
         "fused bilinear RandomResizedCrop + flip + normalize");
   m.def("crop_translate", &turboprune::crop_translate,
