@@ -11,13 +11,14 @@ import torch
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 from turboprune_amd.ops._backend import extension  # noqa: E402
 
+BS = int(os.environ.get("CONV_BS", "32"))
 SHAPES = [  # (N, Cin, H, W, Cout, k, stride, pad)
-    (32, 64, 56, 56, 64, 3, 1, 1),
-    (32, 128, 28, 28, 128, 3, 1, 1),
-    (32, 256, 14, 14, 256, 3, 1, 1),
-    (32, 512, 7, 7, 512, 3, 1, 1),
-    (32, 128, 56, 56, 128, 3, 2, 1),   # stride-2
-    (32, 64, 56, 56, 256, 1, 1, 0),    # 1x1 via the same kernel
+    (BS, 64, 56, 56, 64, 3, 1, 1),
+    (BS, 128, 28, 28, 128, 3, 1, 1),
+    (BS, 256, 14, 14, 256, 3, 1, 1),
+    (BS, 512, 7, 7, 512, 3, 1, 1),
+    (BS, 128, 56, 56, 128, 3, 2, 1),   # stride-2
+    (BS, 64, 56, 56, 256, 1, 1, 0),    # 1x1 via the same kernel
 ]
 
 

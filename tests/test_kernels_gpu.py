@@ -387,3 +387,25 @@ def test_random_resized_crop_gpu_matches_oracle(ext):
     got = A.random_resized_crop(imgs.to(DEV), boxes.to(DEV), mean.to(DEV),
                                 std.to(DEV), 64, flip.to(DEV))
     assert (got.cpu() - ref).abs().max().item() < 1e-4
+
+
+# ------------------------------------------------------- implicit conv
+@pytest.mark.parametrize("shape", [
+    (8, 64, 28, 28, 64, 3, 1, 1),
+    (8, 128, 14, 14, 256, 3, 2, 1),
+    (8, 64, 16, 16, 128, 1, 1, 0),
+])
+def test_conv2d_implicit_fwd_matches_miopen(ext, shape):
+    N, Cin, H, W, Cout, k, s, p = shape
+    torch.manual_seed(Cin + Cout)
+    x = (torch.rand(N, Cin, H, W, device=DEV) - 0.5).to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last)
+    w = ((torch.rand(Cout, Cin, k, k, device=DEV) - 0.5) * 0.1) \
+        .to(torch.bfloat16).to(memory_format=torch.channels_last)
+    b = torch.randn(Cout, device=DEV).to(torch.bfloat16)
+    y = ext.conv2d_implicit_fwd(x, w, b, s, p)
+    ref = torch.nn.functional.conv2d(x, w, b, s, p)
+    assert y.is_contiguous(memory_format=torch.channels_last)
+    scale = ref.float().abs().max().item()
+    assert (y.float() - ref.float()).abs().max().item() < \
+        0.05 * max(scale, 1.0)
