@@ -59,9 +59,6 @@ def main():
             "miopen_TF": round(flop / t_miopen / 1e6, 1)}))
 
 
-if __name__ == "__main__":
-    main()
-    validate_grad_input()
 
 
 def grad_input_via_fwd(ext, gy, w, stride, pad):
@@ -100,3 +97,8 @@ def validate_grad_input():
                           "ok": err < 0.05 * max(scale, 1.0),
                           "ours_us": round(t_ours, 1),
                           "autograd_us": round(t_mi, 1)}))
+
+
+if __name__ == "__main__":
+    main()
+    validate_grad_input()
