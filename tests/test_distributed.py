@@ -154,3 +154,59 @@ def test_ddp_masked_train_step_matches_single_process():
         loss.backward()
         opt.step()
     assert torch.allclose(weights[0], model[0].weight.detach(), atol=1e-5)
+
+
+def _worker_driver(rank, world, port, q):
+    """Full run_experiment flow on 2 gloo ranks: ImageNet-synthetic,
+    ER-ERK prune on rank 0 -> broadcast -> DDP epoch -> checkpoints."""
+    import tempfile
+
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    try:
+        from run_experiment import run
+        from turboprune_amd.config import compose
+
+        tmp = tempfile.mkdtemp(prefix=f"tp_ddp_{rank}_")
+        cfg = compose("bench_resnet50_imagenet", [
+            "model_params.model_name=resnet18",
+            "experiment_params.epochs_per_level=1",
+            "experiment_params.distributed=true",
+            "dataset_params.total_batch_size=8",
+            "+dataset_params.steps_per_epoch=2",
+            f"experiment_params.base_dir={tmp}/experiments",
+            f"dataset_params.data_root_dir={tmp}/data",
+            "pruning_params=pai_er_erk",
+            "pruning_params.target_sparsity=0.5",
+        ])
+        expt_dir = run(cfg)
+        q.put(("ok", rank, expt_dir if rank == 0 else ""))
+    except BaseException as e:  # noqa: BLE001
+        q.put(("err", rank, repr(e)))
+        raise
+    finally:
+        for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR",
+                  "MASTER_PORT"):
+            os.environ.pop(k, None)
+
+
+@pytest.mark.timeout(600)
+def test_run_experiment_distributed_two_ranks():
+    results = _run_workers(_worker_driver)
+    states = {rank: (tag, payload) for (tag, rank, payload) in results}
+    assert states[0][0] == "ok", states[0][1]
+    assert states[1][0] == "ok", states[1][1]
+    expt_dir = states[0][1]
+    assert os.path.exists(os.path.join(expt_dir, "checkpoints",
+                                       "model_level_0.pt"))
+    sd = torch.load(os.path.join(expt_dir, "checkpoints",
+                                 "model_level_0.pt"), weights_only=True)
+    masks = [v for k, v in sd.items() if k.endswith("mask")]
+    total = sum(v.numel() for v in masks)
+    zeros = sum(int((v == 0).sum()) for v in masks)
+    # ER-ERK clamping makes realized sparsity exceed the request
+    # (reference formula, see test_pruning) — just require a real prune
+    assert 0.4 < zeros / total < 0.75
