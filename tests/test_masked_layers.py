@@ -119,3 +119,20 @@ def test_sparsity_accounting():
     layer.mask.zero_()
     layer.mask.view(-1)[: layer.mask.numel() // 4].fill_(1)
     assert abs(layer.sparsity() - 0.75) < 1e-6
+
+
+def test_cache_staleness_guard_on_foreign_mutation():
+    """A dispatcher-level weight mutation (foreign optimizer, manual
+    add_) must invalidate the masked-weight cache automatically."""
+    torch.manual_seed(9)
+    layer = LinearMask(in_features=6, out_features=3, bias=False)
+    layer.mask.bernoulli_(0.5)
+    layer.enable_cache(torch.float32)
+    x = torch.randn(2, 6)
+    y0 = layer(x)
+    with torch.no_grad():
+        layer.weight.add_(1.0)  # bumps _version; cache NOT refreshed here
+    y1 = layer(x)
+    ref = torch.nn.functional.linear(x, layer.mask * layer.weight)
+    assert torch.allclose(y1, ref)
+    assert not torch.allclose(y0, y1)

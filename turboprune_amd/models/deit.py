@@ -54,7 +54,7 @@ class PatchEmbed(nn.Module):
                    .reshape(B, (H // P) * (W // P), P * P * C))
             w2 = self.proj.weight.permute(0, 2, 3, 1).reshape(E, -1)
             m2 = self.proj.mask.permute(0, 2, 3, 1).reshape(E, -1)
-            cache = self.proj.weight_masked
+            cache = self.proj._fresh_cache()
             c2 = cache.permute(0, 2, 3, 1).reshape(E, -1) \
                 if cache is not None else None
             return TF.masked_linear(xp, w2, m2, self.proj.bias, c2,

@@ -43,6 +43,7 @@ class _MaskedMixin:
         # never auto-moved (recreated on device by enable_cache/refresh_cache)
         self.weight_masked: Optional[torch.Tensor] = None
         self.compute_dtype: Optional[torch.dtype] = None
+        self._w_version = -1  # staleness guard (see _fresh_cache)
 
     def set_er_mask(self, p: float, seed: Optional[int] = None) -> None:
         TF.bernoulli_mask_(self.mask, float(p), seed)
@@ -58,9 +59,23 @@ class _MaskedMixin:
         self.weight_masked = None
         self.compute_dtype = None
 
+    def _fresh_cache(self) -> Optional[torch.Tensor]:
+        """The cache, guaranteed fresh. The fused SGD kernel writes weight
+        data in place WITHOUT bumping the tensor version (and rewrites the
+        cache itself in the same pass), so an unchanged `_version` means
+        the cache is valid; any dispatcher-level mutation (a foreign
+        optimizer's add_, copy_, load) bumps it and triggers a refresh
+        here instead of silently serving stale compute weights."""
+        if self.weight_masked is None:
+            return None
+        if self.weight._version != self._w_version:
+            self.refresh_cache()
+        return self.weight_masked
+
     def refresh_cache(self) -> None:
         if self.compute_dtype is None:
             return
+        self._w_version = self.weight._version
         with torch.no_grad():
             wm = TF.mask_apply(self.weight, self.mask, self.compute_dtype)
         if self.weight_masked is not None and \
@@ -113,7 +128,7 @@ class ConvMask(_MaskedMixin, nn.Conv2d):
         if self._gemm_1x1_ok(x):
             n, c, h, w_ = x.shape
             x2 = x.permute(0, 2, 3, 1).reshape(-1, c)  # free view on NHWC
-            cache = self.weight_masked
+            cache = self._fresh_cache()
             y2 = TF.masked_linear(
                 x2, self.weight.reshape(self.out_channels, c),
                 self.mask.reshape(self.out_channels, c), self.bias,
@@ -121,7 +136,7 @@ class ConvMask(_MaskedMixin, nn.Conv2d):
                 if cache is not None else None,
                 self.compute_dtype)
             return y2.view(n, h, w_, self.out_channels).permute(0, 3, 1, 2)
-        w = TF.masked_weight(self.weight, self.mask, self.weight_masked,
+        w = TF.masked_weight(self.weight, self.mask, self._fresh_cache(),
                              self.compute_dtype)
         return torch.nn.functional.conv2d(
             x, w, _bias_like(self.bias, w), self.stride, self.padding,
@@ -138,7 +153,7 @@ class LinearMask(_MaskedMixin, nn.Linear):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return TF.masked_linear(x, self.weight, self.mask, self.bias,
-                                self.weight_masked, self.compute_dtype)
+                                self._fresh_cache(), self.compute_dtype)
 
 
 class Conv1dMask(_MaskedMixin, nn.Conv1d):
@@ -154,7 +169,7 @@ class Conv1dMask(_MaskedMixin, nn.Conv1d):
         self._init_mask()
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        cache = self.weight_masked
+        cache = self._fresh_cache()
         return TF.masked_linear(
             x, self.weight.squeeze(-1), self.mask.squeeze(-1), self.bias,
             cache.squeeze(-1) if cache is not None else None,
