@@ -99,13 +99,14 @@ def prune_snip(model: nn.Module, density: float,
 
 def prune_synflow(model: nn.Module, density: float,
                   dataloader: Iterable,
-                  device: Optional[torch.device] = None) -> nn.Module:
+                  device: Optional[torch.device] = None,
+                  amp_dtype: torch.dtype = torch.bfloat16) -> nn.Module:
     """SynFlow: linearize params to |param| (signs saved), backward of
     sum(model(ones_input)), score = |mask * grad * weight|, restore signs
-    (reference: pruning_utils.py:208-285)."""
+    (reference: pruning_utils.py:208-285). The forward runs in the
+    model's CURRENT train/eval mode and under autocast, exactly as the
+    reference does (BN batch stats / running-stat updates included)."""
     device = device or _model_device(model)
-    was_training = model.training
-    model.eval()  # BN in eval so the all-ones pass is well defined
 
     # linearize: theta <- |theta| over ALL params & buffers, keep signs
     signs = {}
@@ -122,10 +123,13 @@ def prune_synflow(model: nn.Module, density: float,
     ones = torch.ones(input_shape, device=device)
 
     model.zero_grad(set_to_none=True)
-    out = model(ones)
-    if isinstance(out, tuple):
-        out = out[0]
-    torch.sum(out).backward()
+    use_amp = device.type == "cuda" and amp_dtype != torch.float32
+    with torch.autocast(device_type="cuda", dtype=amp_dtype,
+                        enabled=use_amp):
+        out = model(ones)
+        if isinstance(out, tuple):
+            out = out[0]
+        torch.sum(out).backward()
 
     with torch.no_grad():
         scores = {}
@@ -143,7 +147,6 @@ def prune_synflow(model: nn.Module, density: float,
         if thr is not None:
             _rewrite_masks(model, scores, thr)
     model.zero_grad(set_to_none=True)
-    model.train(was_training)
     return model
 
 
@@ -245,7 +248,11 @@ def prune_the_model(cfg: Any, model: nn.Module, target_density: float,
         return prune_snip(model, target_density, dataloader, device, amp)
     if method == "synflow":
         assert dataloader is not None, "synflow needs a train dataloader"
-        return prune_synflow(model, target_density, dataloader, device)
+        amp = {"bfloat16": torch.bfloat16, "float16": torch.float16,
+               "float32": torch.float32}[
+                   cfg.select("experiment_params.training_precision",
+                              "bfloat16")]
+        return prune_synflow(model, target_density, dataloader, device, amp)
     if method == "random_erk":
         return prune_random_erk(model, target_density)
     if method == "random_balanced":
