@@ -32,6 +32,7 @@ sources = [
     os.path.join(CSRC, "layernorm_gelu.hip"),
     os.path.join(CSRC, "rrc.hip"),
     os.path.join(CSRC, "conv_implicit.hip"),
+    os.path.join(CSRC, "conv_wrw.hip"),
 ]
 
 setup(

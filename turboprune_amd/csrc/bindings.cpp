@@ -30,6 +30,8 @@ at::Tensor transpose2d(const at::Tensor&);
 at::Tensor conv2d_implicit_fwd(const at::Tensor&, const at::Tensor&,
                                const c10::optional<at::Tensor>&, int64_t,
                                int64_t);
+at::Tensor conv2d_implicit_wrw(const at::Tensor&, const at::Tensor&,
+                               int64_t, int64_t, int64_t, int64_t);
 std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_fwd(
     const at::Tensor&, const at::Tensor&, const at::Tensor&, double);
 std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_bwd(
@@ -87,6 +89,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gelu_bwd", &turboprune::gelu_bwd, "fused exact GELU bwd");
   m.def("conv2d_implicit_fwd", &turboprune::conv2d_implicit_fwd,
         "implicit-GEMM NHWC conv forward (experimental)");
+  m.def("conv2d_implicit_wrw", &turboprune::conv2d_implicit_wrw,
+        "implicit-GEMM NHWC conv weight gradient (experimental)");
   m.def("random_resized_crop", &turboprune::random_resized_crop,
         "fused bilinear RandomResizedCrop + flip + normalize");
   m.def("crop_translate", &turboprune::crop_translate,

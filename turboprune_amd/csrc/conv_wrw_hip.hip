@@ -1,0 +1,245 @@
+#include "hip/hip_runtime.h"
+// Implicit-GEMM conv WEIGHT gradient (wrw) on MFMA — experimental.
+//
+//   gw[co][dh,dw,ci] = sum_opix gy[opix][co] * x[n, ho*s-p+dh, wo*s-p+dw, ci]
+//
+// A TN GEMM: both operands are M-major over output pixels (the deep
+// contraction, N*Ho*Wo up to ~1.6M), so tiles are loaded 16B-coalesced
+// along their fast dims (co / ci) and TRANSPOSED into K-major LDS images
+// by ds_write scatter (glds cannot transpose — guide §5.4 rule 21).
+// Split-K over opix chunks (gridDim.y slabs -> fp32 partials -> the
+// shared splitk reduce), since the output (Cout x 9Cin) is tiny.
+//
+// Tile: C 128(co) x 128(tapci), BK = 64 output pixels. LDS images
+// [row][k] like gemm_bt (XOR-swizzled 16B blocks), so the MFMA fragment
+// reads are identical to gemm_bt's.
+
+#include <ATen/ATen.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace turboprune {
+
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+namespace conv_wrw {
+constexpr int BM = 128, BN = 128, BK = 64;
+constexpr int WARPS_N = 2;
+constexpr int WM = 64, WN = 64;
+constexpr int MREP = 4, NREP = 4;
+
+TP_DEVICE int lds_byte(int row, int k) {
+  int blk = (k >> 3) ^ (row & 7);
+  return row * (BK * 2) + blk * 16 + (k & 7) * 2;
+}
+}  // namespace conv_wrw
+
+// one slab computes K-tiles [kt0, kt1) of the opix axis into
+// partial[blockIdx.y * (Mp*Np) + ...] (fp32)
+__global__ __launch_bounds__(256) void conv_wrw_kernel(
+    const __hip_bfloat16* __restrict__ gy,  // (M, Cout) row-major
+    const __hip_bfloat16* __restrict__ x,   // NHWC
+    float* __restrict__ partial,            // (slabs, Mp, Np)
+    int Nb, int Hi, int Wi, int Cin, int Cout, int Ho, int Wo, int KH,
+    int KW, int stride, int pad, int Mp, int Np, int grid_n) {
+  using namespace conv_wrw;
+  // LDS: 2 x (A,B) tiles of 128x64 bf16 = 64 KiB (single-buffered pairs,
+  // one barrier per K-tile; staging is register->ds_write)
+  __shared__ char smem[2 * BM * BK * 2];
+  char* sA = smem;                     // [co][opix]
+  char* sB = smem + BM * BK * 2;       // [tapci][opix]
+
+  int wg = blockIdx.x;
+  int tile_m = (wg / grid_n) * BM;     // co tile
+  int tile_n = (wg % grid_n) * BN;     // tapci tile
+  int64_t M = (int64_t)Nb * Ho * Wo;
+
+  int lane = threadIdx.x & (kWave - 1);
+  int wid = threadIdx.x / kWave;
+  int wr = wid / WARPS_N, wc = wid % WARPS_N;
+
+  int total_kt = (int)((M + BK - 1) / BK);
+  int per = (total_kt + gridDim.y - 1) / gridDim.y;
+  int kt0 = blockIdx.y * per;
+  int kt1 = min(kt0 + per, total_kt);
+  partial += (int64_t)blockIdx.y * Mp * Np;
+
+  f32x4 acc[MREP][NREP] = {};
+
+  for (int kt = kt0; kt < kt1; ++kt) {
+    int64_t opix0 = (int64_t)kt * BK;
+    // ---- stage A: gy^T tile [128 co][64 opix] -------------------------
+    // 256 threads x 2 iters: each thread loads 8 co (16B) of one opix,
+    // then scatters 8 ds_write_b16 into [co][opix].
+    {
+      int op = threadIdx.x & 63;        // opix within tile
+      int co8 = threadIdx.x >> 6;       // 0..3
+      // 4 iterations x 4 co8 = 16 co-octets = all 128 rows
+#pragma unroll
+      for (int it = 0; it < 4; ++it) {
+        int co0 = (co8 + it * 4) * 8;
+        int64_t opix = opix0 + op;
+        __hip_bfloat16 vals[8];
+        if (opix < M && tile_m + co0 < Cout) {
+          const __hip_bfloat16* src = gy + opix * Cout + tile_m + co0;
+          *reinterpret_cast<uint4*>(vals) =
+              *reinterpret_cast<const uint4*>(src);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.f);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int row = co0 + j;
+          *reinterpret_cast<__hip_bfloat16*>(sA + lds_byte(row, op)) =
+              vals[j];
+        }
+      }
+    }
+    // ---- stage B: im2col^T tile [128 tapci][64 opix] ------------------
+    {
+      int op = threadIdx.x & 63;
+      int tc8 = threadIdx.x >> 6;
+#pragma unroll
+      for (int it = 0; it < 4; ++it) {
+        int tc0 = (tc8 + it * 4) * 8;   // tapci octet start
+        int64_t opix = opix0 + op;
+        __hip_bfloat16 vals[8];
+        bool ok = false;
+        if (opix < M && tile_n + tc0 < KH * KW * Cin) {
+          int tapci = tile_n + tc0;
+          int tap = tapci / Cin;        // Cin % 8 == 0 keeps octet in tap
+          int ci = tapci % Cin;
+          int dh = tap / KW, dw = tap % KW;
+          int wo = (int)(opix % Wo);
+          int64_t r2 = opix / Wo;
+          int ho = (int)(r2 % Ho);
+          int n = (int)(r2 / Ho);
+          int hi = ho * stride - pad + dh;
+          int wi = wo * stride - pad + dw;
+          if (hi >= 0 && hi < Hi && wi >= 0 && wi < Wi) {
+            const __hip_bfloat16* src =
+                x + (((int64_t)n * Hi + hi) * Wi + wi) * Cin + ci;
+            *reinterpret_cast<uint4*>(vals) =
+                *reinterpret_cast<const uint4*>(src);
+            ok = true;
+          }
+        }
+        if (!ok) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) vals[j] = __float2bfloat16(0.f);
+        }
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          *reinterpret_cast<__hip_bfloat16*>(sB + lds_byte(tc0 + j, op)) =
+              vals[j];
+        }
+      }
+    }
+    __syncthreads();
+    // ---- MFMA over the 64-opix tile ----------------------------------
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 a_frag[MREP], b_frag[NREP];
+      int kf = ks * 32 + (lane >> 4) * 8;
+      int rowf = lane & 15;
+#pragma unroll
+      for (int mi = 0; mi < MREP; ++mi)
+        a_frag[mi] = *reinterpret_cast<const bf16x8*>(
+            sA + lds_byte(wr * WM + mi * 16 + rowf, kf));
+#pragma unroll
+      for (int ni = 0; ni < NREP; ++ni)
+        b_frag[ni] = *reinterpret_cast<const bf16x8*>(
+            sB + lds_byte(wc * WN + ni * 16 + rowf, kf));
+#pragma unroll
+      for (int mi = 0; mi < MREP; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NREP; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // write the fp32 slab (direct; Mp/Np padded so no guards)
+#pragma unroll
+  for (int mi = 0; mi < MREP; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < NREP; ++ni) {
+      int col = tile_n + wc * WN + ni * 16 + (lane & 15);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = tile_m + wr * WM + mi * 16 + (lane >> 4) * 4 + j;
+        partial[(int64_t)row * Np + col] = acc[mi][ni][j];
+      }
+    }
+  }
+}
+
+__global__ void wrw_reduce_kernel(const float* __restrict__ partial,
+                                  __hip_bfloat16* __restrict__ out,
+                                  int64_t mn, int64_t slab_stride,
+                                  int slabs, int Np, int N_real,
+                                  int64_t out_cols) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < mn;
+       i += stride) {
+    int64_t row = i / out_cols;
+    int64_t col = i % out_cols;
+    float v = 0.f;
+    for (int s = 0; s < slabs; ++s)
+      v += partial[(int64_t)s * slab_stride + row * Np + col];
+    out[i] = __float2bfloat16(v);
+  }
+}
+
+// gy: (N, Cout, Ho, Wo) channels_last; x: (N, Cin, Hi, Wi) channels_last.
+// Returns grad_weight (Cout, Cin, KH, KW) channels_last bf16.
+at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
+                               const at::Tensor& x_in, int64_t KH,
+                               int64_t KW, int64_t stride, int64_t pad) {
+  using namespace conv_wrw;
+  auto gy = gy_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
+  int Nb = x.size(0), Cin = x.size(1), Hi = x.size(2), Wi = x.size(3);
+  int Cout = gy.size(1), Ho = gy.size(2), Wo = gy.size(3);
+  TORCH_CHECK(Cin % 8 == 0, "wrw: Cin % 8 != 0");
+  int64_t M = (int64_t)Nb * Ho * Wo;
+  int K = (int)(KH * KW) * Cin;
+  int Mp = (Cout + BM - 1) / BM * BM;
+  int Np = (K + BN - 1) / BN * BN;
+
+  int grid_m = Mp / BM, grid_n = Np / BN;
+  int tiles = grid_m * grid_n;
+  int total_kt = (int)((M + BK - 1) / BK);
+  int splitk = 1;
+  while (tiles * splitk < 384 && splitk * 4 <= total_kt && splitk < 64)
+    splitk *= 2;
+
+  auto partial = at::empty({splitk, (int64_t)Mp, (int64_t)Np},
+                           x.options().dtype(at::kFloat));
+  auto stream = at::hip::getCurrentHIPStream();
+  dim3 grid(tiles, splitk);
+  hipLaunchKernelGGL(conv_wrw_kernel, grid, dim3(256), 0, stream,
+                     reinterpret_cast<const __hip_bfloat16*>(gy.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                     partial.data_ptr<float>(), Nb, Hi, Wi, Cin, Cout, Ho,
+                     Wo, (int)KH, (int)KW, (int)stride, (int)pad, Mp, Np,
+                     grid_n);
+
+  // reduce slabs -> (Cout, K) bf16, channels_last weight memory order
+  auto gw = at::empty({Cout, Cin, KH, KW},
+                      gy.options().memory_format(
+                          at::MemoryFormat::ChannelsLast));
+  int64_t mn = (int64_t)Cout * K;
+  int rgrid = elementwise_grid(mn, kBlock, 4);
+  hipLaunchKernelGGL(wrw_reduce_kernel, dim3(rgrid), dim3(kBlock), 0,
+                     stream, partial.data_ptr<float>(),
+                     reinterpret_cast<__hip_bfloat16*>(gw.data_ptr()), mn,
+                     (int64_t)Mp * Np, splitk, Np, K, (int64_t)K);
+  return gw;
+}
+
+}  // namespace turboprune
