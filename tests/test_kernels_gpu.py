@@ -409,3 +409,25 @@ def test_conv2d_implicit_fwd_matches_miopen(ext, shape):
     scale = ref.float().abs().max().item()
     assert (y.float() - ref.float()).abs().max().item() < \
         0.05 * max(scale, 1.0)
+
+
+@pytest.mark.parametrize("shape", [
+    (8, 64, 28, 28, 64, 3, 1, 1),
+    (8, 64, 28, 28, 128, 3, 2, 1),
+    (4, 64, 16, 16, 128, 1, 1, 0),
+])
+def test_conv2d_implicit_wrw_matches_autograd(ext, shape):
+    N, Cin, H, W, Cout, k, s, p = shape
+    torch.manual_seed(Cin + Cout + s)
+    x = (torch.rand(N, Cin, H, W, device=DEV) - 0.5).to(torch.bfloat16) \
+        .to(memory_format=torch.channels_last).requires_grad_()
+    w = ((torch.rand(Cout, Cin, k, k, device=DEV) - 0.5) * 0.1) \
+        .to(torch.bfloat16).to(memory_format=torch.channels_last) \
+        .requires_grad_()
+    y = torch.nn.functional.conv2d(x, w, None, s, p)
+    gy = torch.randn_like(y)
+    (ref,) = torch.autograd.grad(y, w, gy)
+    got = ext.conv2d_implicit_wrw(gy, x.detach(), k, k, s, p)
+    scale = ref.float().abs().max().item()
+    assert (got.float() - ref.float()).abs().max().item() < \
+        0.05 * max(scale, 1.0)
