@@ -59,10 +59,14 @@ def _worker_ddp_parity(rank, world, port, q):
     dist.destroy_process_group()
 
 
+_PORT_SALT = [0]
+
+
 def _run_workers(fn, world=2):
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    port = 29531 + (os.getpid() % 500)
+    _PORT_SALT[0] += 7  # distinct port per call: no TIME_WAIT collisions
+    port = 29531 + (os.getpid() % 500) + _PORT_SALT[0]
     procs = [ctx.Process(target=fn, args=(r, world, port, q))
              for r in range(world)]
     for p in procs:
