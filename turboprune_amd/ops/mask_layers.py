@@ -138,6 +138,12 @@ class ConvMask(_MaskedMixin, nn.Conv2d):
             return y2.view(n, h, w_, self.out_channels).permute(0, 3, 1, 2)
         w = TF.masked_weight(self.weight, self.mask, self._fresh_cache(),
                              self.compute_dtype)
+        from turboprune_amd.ops import conv_native
+        if conv_native.native_conv_ok(x, w, self.stride, self.padding,
+                                      self.dilation, self.groups):
+            return conv_native.conv2d(
+                x, w, _bias_like(self.bias, w), self.stride[0],
+                self.padding[0], conv_native.NativeBackend)
         return torch.nn.functional.conv2d(
             x, w, _bias_like(self.bias, w), self.stride, self.padding,
             self.dilation, self.groups)
