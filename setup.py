@@ -33,6 +33,7 @@ sources = [
     os.path.join(CSRC, "rrc.hip"),
     os.path.join(CSRC, "conv_implicit.hip"),
     os.path.join(CSRC, "conv_wrw.hip"),
+    os.path.join(CSRC, "gemm_256_8phase.hip"),
 ]
 
 setup(

@@ -25,6 +25,8 @@ std::tuple<at::Tensor, at::Tensor> linear_bwd(const at::Tensor&,
                                               const at::Tensor&,
                                               const at::Tensor&);
 at::Tensor gemm_bf16(const at::Tensor&, const at::Tensor&, bool, bool);
+at::Tensor gemm_bt_256(const at::Tensor&, const at::Tensor&,
+                       const c10::optional<at::Tensor>&, bool);
 // fused NHWC batchnorm (batchnorm.hip)
 at::Tensor transpose2d(const at::Tensor&);
 at::Tensor conv2d_implicit_fwd(const at::Tensor&, const at::Tensor&,
@@ -83,6 +85,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "MFMA GEMM grads (grad_x, grad_w)");
   m.def("gemm_bf16", &turboprune::gemm_bf16,
         "raw MFMA bf16 GEMM (testing entry)");
+  m.def("gemm_bt_256", &turboprune::gemm_bt_256,
+        "EXPERIMENTAL 256x256 8-phase MFMA GEMM (unwired; needs "
+        "on-device race screen before promotion)");
   m.def("ln_fwd", &turboprune::ln_fwd, "fused LayerNorm fwd");
   m.def("ln_bwd", &turboprune::ln_bwd, "fused LayerNorm bwd");
   m.def("gelu_fwd", &turboprune::gelu_fwd, "fused exact GELU fwd");
