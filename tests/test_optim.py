@@ -114,3 +114,45 @@ def test_schedule_free_decreases_loss():
         if first is None:
             first = loss.item()
     assert loss.item() < first * 0.2
+
+
+def test_multi_sgd_env_flag_cpu_fallback(monkeypatch):
+    """TURBOPRUNE_MULTI_SGD=1 must not change CPU behavior (the multi
+    path is GPU-only); parity vs torch.optim.SGD still holds."""
+    import torch
+
+    from turboprune_amd.ops.mask_layers import LinearMask
+    from turboprune_amd.optim import FusedMaskedSGD
+    monkeypatch.setenv("TURBOPRUNE_MULTI_SGD", "1")
+    torch.manual_seed(11)
+    m = LinearMask(in_features=6, out_features=4, bias=True)
+    m.mask.bernoulli_(0.5)
+    ref = torch.nn.Linear(6, 4)
+    ref.load_state_dict({"weight": m.weight.detach().clone(),
+                         "bias": m.bias.detach().clone()})
+    opt = FusedMaskedSGD(m.parameters(), lr=0.1, momentum=0.9,
+                         weight_decay=1e-4, model=m)
+    ropt = torch.optim.SGD(ref.parameters(), lr=0.1, momentum=0.9,
+                           weight_decay=1e-4)
+    for _ in range(3):
+        g = torch.randn_like(m.weight)
+        m.weight.grad = g.clone()
+        m.bias.grad = torch.randn_like(m.bias)
+        ref.weight.grad = g.clone()
+        ref.bias.grad = m.bias.grad.clone()
+        opt.step()
+        ropt.step()
+    assert torch.allclose(m.weight, ref.weight, atol=1e-6)
+
+
+def test_multi_ok_layout_guard():
+    import torch
+
+    from turboprune_amd.optim.sgd import FusedMaskedSGD
+    p = torch.randn(4, 3, 3, 3)
+    good = torch.randn_like(p)
+    bad = torch.randn(4, 3, 3, 3).to(memory_format=torch.channels_last)
+    assert FusedMaskedSGD._multi_ok(p, good, None, None, None)
+    assert not FusedMaskedSGD._multi_ok(p, bad, None, None, None)
+    assert not FusedMaskedSGD._multi_ok(p.to(torch.bfloat16), good, None,
+                                        None, None)

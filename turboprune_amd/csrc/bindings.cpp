@@ -27,6 +27,9 @@ std::tuple<at::Tensor, at::Tensor> linear_bwd(const at::Tensor&,
 at::Tensor gemm_bf16(const at::Tensor&, const at::Tensor&, bool, bool);
 at::Tensor gemm_bt_256(const at::Tensor&, const at::Tensor&,
                        const c10::optional<at::Tensor>&, bool);
+void sgd_step_multi_(std::vector<at::Tensor>, std::vector<at::Tensor>,
+                     std::vector<at::Tensor>, std::vector<at::Tensor>,
+                     std::vector<at::Tensor>, double, double, double);
 // fused NHWC batchnorm (batchnorm.hip)
 at::Tensor transpose2d(const at::Tensor&);
 at::Tensor conv2d_implicit_fwd(const at::Tensor&, const at::Tensor&,
@@ -71,6 +74,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "Philox Bernoulli(p) mask fill");
   m.def("sgd_step_", &turboprune::sgd_step_,
         "fused SGD momentum+wd step with mask-reapply cache rewrite");
+  m.def("sgd_step_multi_", &turboprune::sgd_step_multi_,
+        "multi-tensor fused SGD step (one launch over many params; "
+        "uniform flags per call — opt-in via TURBOPRUNE_MULTI_SGD=1)");
   m.def("kth_smallest", &turboprune::kth_smallest,
         "radix-select k-th smallest of a 1-D fp32 tensor");
   m.def("ce_fwd", &turboprune::ce_fwd, "fused CE forward -> (loss, lse)");

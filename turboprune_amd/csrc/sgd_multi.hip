@@ -1,0 +1,161 @@
+// Multi-tensor fused SGD step (horizontal fusion of sgd_fused.hip).
+//
+// One launch sweeps MANY parameters: a ResNet50 step otherwise issues
+// ~161 back-to-back sgd_step_ launches; at small per-GPU batches
+// (strong scaling) that launch train is a visible slice of the step.
+// Descriptor-table design (apex-style): a [n,6] int64 table of device
+// pointers {w, g, buf, mask, cache, numel} plus a [blocks,2] int32
+// block->(tensor, chunk) map, both built host-side per call and
+// uploaded once (~8 KB). Flags (momentum/mask/cache/grad dtype) are
+// UNIFORM per call — the Python side groups parameters accordingly.
+//
+// The math and the flat-in-weight-storage-order contract are identical
+// to sgd_step_ (see sgd_fused.hip): every operand must already share
+// the weight's dense strides (the Python grouping falls back to the
+// per-tensor path for any parameter whose operands mismatch).
+
+#include <ATen/ATen.h>
+#include <ATen/hip/HIPContext.h>
+
+#include <vector>
+
+#include "common.h"
+
+namespace turboprune {
+
+namespace {
+constexpr int64_t kChunk = 1 << 16;  // elements per block
+}
+
+template <typename GradT, typename CacheT, bool kMomentum, bool kMask,
+          bool kCache>
+__global__ void sgd_multi_kernel(const int64_t* __restrict__ desc,
+                                 const int* __restrict__ blockmap,
+                                 float lr, float momentum, float wd) {
+  int t = blockmap[blockIdx.x * 2];
+  int c = blockmap[blockIdx.x * 2 + 1];
+  const int64_t* d = desc + (int64_t)t * 6;
+  float* __restrict__ w = reinterpret_cast<float*>(d[0]);
+  const GradT* __restrict__ g = reinterpret_cast<const GradT*>(d[1]);
+  float* __restrict__ buf = reinterpret_cast<float*>(d[2]);
+  const float* __restrict__ mask = reinterpret_cast<const float*>(d[3]);
+  CacheT* __restrict__ cache = reinterpret_cast<CacheT*>(d[4]);
+  int64_t n = d[5];
+  int64_t end = min((int64_t)(c + 1) * kChunk, n);
+  for (int64_t i = (int64_t)c * kChunk + threadIdx.x; i < end;
+       i += blockDim.x) {
+    float wi = w[i];
+    float dv = to_float<GradT>(g[i]) + wd * wi;
+    if (kMomentum) {
+      float b = momentum * buf[i] + dv;
+      buf[i] = b;
+      dv = b;
+    }
+    wi -= lr * dv;
+    w[i] = wi;
+    if (kCache) {
+      float cv = kMask ? (mask[i] != 0.f ? wi : 0.f) : wi;
+      cache[i] = from_float<CacheT>(cv);
+    }
+  }
+}
+
+void sgd_step_multi_(std::vector<at::Tensor> ws, std::vector<at::Tensor> gs,
+                     std::vector<at::Tensor> bufs,
+                     std::vector<at::Tensor> masks,
+                     std::vector<at::Tensor> caches, double lr,
+                     double momentum, double wd) {
+  size_t n = ws.size();
+  TORCH_CHECK(n > 0 && gs.size() == n, "sgd_step_multi_: empty/mismatched");
+  bool has_m = momentum != 0.0 && bufs.size() == n;
+  bool has_mask = masks.size() == n;
+  bool has_cache = caches.size() == n;
+  auto grad_t = gs[0].scalar_type();
+  auto cache_t = has_cache ? caches[0].scalar_type() : at::kFloat;
+
+  auto desc_cpu = at::empty({(int64_t)n, 6},
+                            at::TensorOptions().dtype(at::kLong));
+  int64_t* dp = desc_cpu.data_ptr<int64_t>();
+  std::vector<int> bm;
+  bm.reserve(1024);
+  for (size_t i = 0; i < n; ++i) {
+    auto& w = ws[i];
+    TORCH_CHECK(w.is_cuda() && w.scalar_type() == at::kFloat &&
+                w.is_non_overlapping_and_dense(),
+                "sgd_step_multi_: weight must be dense fp32 on GPU");
+    TORCH_CHECK(gs[i].scalar_type() == grad_t &&
+                gs[i].strides().vec() == w.strides().vec(),
+                "sgd_step_multi_: grad dtype/layout must be uniform & match"
+                " (group on the Python side; fall back per-tensor)");
+    dp[i * 6 + 0] = (int64_t)w.data_ptr();
+    dp[i * 6 + 1] = (int64_t)gs[i].data_ptr();
+    if (has_m) {
+      TORCH_CHECK(bufs[i].scalar_type() == at::kFloat &&
+                  bufs[i].strides().vec() == w.strides().vec());
+      dp[i * 6 + 2] = (int64_t)bufs[i].data_ptr();
+    } else {
+      dp[i * 6 + 2] = 0;
+    }
+    if (has_mask) {
+      TORCH_CHECK(masks[i].scalar_type() == at::kFloat &&
+                  masks[i].strides().vec() == w.strides().vec());
+      dp[i * 6 + 3] = (int64_t)masks[i].data_ptr();
+    } else {
+      dp[i * 6 + 3] = 0;
+    }
+    if (has_cache) {
+      TORCH_CHECK(caches[i].scalar_type() == cache_t &&
+                  caches[i].strides().vec() == w.strides().vec());
+      dp[i * 6 + 4] = (int64_t)caches[i].data_ptr();
+    } else {
+      dp[i * 6 + 4] = 0;
+    }
+    int64_t numel = w.numel();
+    dp[i * 6 + 5] = numel;
+    int chunks = (int)((numel + kChunk - 1) / kChunk);
+    for (int c = 0; c < chunks; ++c) {
+      bm.push_back((int)i);
+      bm.push_back(c);
+    }
+  }
+  int blocks = (int)(bm.size() / 2);
+  auto bm_cpu = at::from_blob(bm.data(), {(int64_t)bm.size()},
+                              at::TensorOptions().dtype(at::kInt)).clone();
+  auto dev = ws[0].device();
+  auto desc = desc_cpu.to(dev);
+  auto bmap = bm_cpu.to(dev);
+  auto stream = at::hip::getCurrentHIPStream();
+
+#define TP_MSGD(GT, CT, MOM, MASKF, CACHEF)                              \
+  hipLaunchKernelGGL((sgd_multi_kernel<GT, CT, MOM, MASKF, CACHEF>),     \
+                     dim3(blocks), dim3(kBlock), 0, stream,              \
+                     desc.data_ptr<int64_t>(), bmap.data_ptr<int>(),     \
+                     (float)lr, (float)momentum, (float)wd)
+#define TP_MSGD_FLAGS(GT, CT)                                            \
+  do {                                                                   \
+    if (has_m) {                                                         \
+      if (has_cache && has_mask) TP_MSGD(GT, CT, true, true, true);      \
+      else if (has_cache) TP_MSGD(GT, CT, true, false, true);            \
+      else TP_MSGD(GT, CT, true, false, false);                          \
+    } else {                                                             \
+      if (has_cache && has_mask) TP_MSGD(GT, CT, false, true, true);     \
+      else if (has_cache) TP_MSGD(GT, CT, false, false, true);           \
+      else TP_MSGD(GT, CT, false, false, false);                         \
+    }                                                                    \
+  } while (0)
+
+  if (grad_t == at::kFloat) {
+    if (cache_t == at::kBFloat16) TP_MSGD_FLAGS(float, __hip_bfloat16);
+    else TP_MSGD_FLAGS(float, float);
+  } else if (grad_t == at::kBFloat16) {
+    if (cache_t == at::kBFloat16)
+      TP_MSGD_FLAGS(__hip_bfloat16, __hip_bfloat16);
+    else TP_MSGD_FLAGS(__hip_bfloat16, float);
+  } else {
+    TORCH_CHECK(false, "sgd_step_multi_: unsupported grad dtype");
+  }
+#undef TP_MSGD_FLAGS
+#undef TP_MSGD
+}
+
+}  // namespace turboprune

@@ -49,17 +49,30 @@ class FusedMaskedSGD(torch.optim.SGD):
         for _, m in masked_modules(model):
             self._masked[id(m.weight)] = m
 
+    @staticmethod
+    def _multi_ok(p, grad, buf, mask, cache) -> bool:
+        """All operands already flat in the weight's storage order — the
+        multi-tensor kernel does no relayout (sgd_multi.hip contract)."""
+        s = p.stride()
+        for t in (grad, buf, mask, cache):
+            if t is not None and (t.stride() != s or t.numel() != p.numel()):
+                return False
+        return p.dtype == torch.float32
+
     @torch.no_grad()
     def step(self, closure=None):  # noqa: C901
         loss = None
         if closure is not None:
             with torch.enable_grad():
                 loss = closure()
+        import os
+        use_multi = os.environ.get("TURBOPRUNE_MULTI_SGD", "") == "1"
 
         for group in self.param_groups:
             lr = group["lr"]
             momentum = group["momentum"]
             wd = group["weight_decay"]
+            buckets: Dict[Any, list] = {}
             for p in group["params"]:
                 if p.grad is None:
                     continue
@@ -77,12 +90,21 @@ class FusedMaskedSGD(torch.optim.SGD):
                 layer = self._masked.get(id(p))
                 cache = getattr(layer, "weight_masked", None) \
                     if layer is not None else None
+                mask = layer.mask if layer is not None else None
 
                 if p.is_cuda and _backend.use_native(p):
                     ext = _backend.extension()
+                    if use_multi and self._multi_ok(p, grad, buf, mask,
+                                                    cache):
+                        key = (grad.dtype, mask is not None,
+                               cache is not None,
+                               cache.dtype if cache is not None else None)
+                        buckets.setdefault(key, []).append(
+                            (p, grad, buf, mask, cache))
+                        continue
                     ext.sgd_step_(
                         p, grad, buf if buf is not None else torch.Tensor(),
-                        layer.mask if layer is not None else torch.Tensor(),
+                        mask if mask is not None else torch.Tensor(),
                         cache if cache is not None else torch.Tensor(),
                         float(lr), float(momentum), float(wd))
                 else:
@@ -95,6 +117,14 @@ class FusedMaskedSGD(torch.optim.SGD):
                     p.add_(d_p, alpha=-lr)
                     if cache is not None and layer is not None:
                         layer.refresh_cache()
+            for (gdt, has_mask, has_cache, _cdt), items in buckets.items():
+                ext = _backend.extension()
+                ext.sgd_step_multi_(
+                    [it[0] for it in items], [it[1] for it in items],
+                    [it[2] for it in items] if momentum != 0 else [],
+                    [it[3] for it in items] if has_mask else [],
+                    [it[4] for it in items] if has_cache else [],
+                    float(lr), float(momentum), float(wd))
         return loss
 
 
