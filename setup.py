@@ -35,6 +35,7 @@ sources = [
     os.path.join(CSRC, "conv_wrw.hip"),
     os.path.join(CSRC, "gemm_256_8phase.hip"),
     os.path.join(CSRC, "sgd_multi.hip"),
+    os.path.join(CSRC, "attention.hip"),
 ]
 
 setup(

@@ -30,6 +30,8 @@ at::Tensor gemm_bt_256(const at::Tensor&, const at::Tensor&,
 void sgd_step_multi_(std::vector<at::Tensor>, std::vector<at::Tensor>,
                      std::vector<at::Tensor>, std::vector<at::Tensor>,
                      std::vector<at::Tensor>, double, double, double);
+at::Tensor attn_fwd(const at::Tensor&, const at::Tensor&,
+                    const at::Tensor&, double);
 // fused NHWC batchnorm (batchnorm.hip)
 at::Tensor transpose2d(const at::Tensor&);
 at::Tensor conv2d_implicit_fwd(const at::Tensor&, const at::Tensor&,
@@ -77,6 +79,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step_multi_", &turboprune::sgd_step_multi_,
         "multi-tensor fused SGD step (one launch over many params; "
         "uniform flags per call — opt-in via TURBOPRUNE_MULTI_SGD=1)");
+  m.def("attn_fwd", &turboprune::attn_fwd,
+        "EXPERIMENTAL fused flash-style attention forward, head_dim 64 "
+        "(unwired; opt-in TURBOPRUNE_ATTN=native after device validation)");
   m.def("kth_smallest", &turboprune::kth_smallest,
         "radix-select k-th smallest of a 1-D fp32 tensor");
   m.def("ce_fwd", &turboprune::ce_fwd, "fused CE forward -> (loss, lse)");

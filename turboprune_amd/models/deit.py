@@ -80,8 +80,13 @@ class Attention(nn.Module):
         qkv = self.qkv(x).reshape(B, N, 3, self.num_heads, self.head_dim)
         qkv = qkv.permute(2, 0, 3, 1, 4)      # 3, B, heads, N, hd
         q, k, v = qkv[0], qkv[1], qkv[2]
-        x = torch.nn.functional.scaled_dot_product_attention(
-            q, k, v, dropout_p=self.attn_drop.p if self.training else 0.0)
+        drop = self.attn_drop.p if self.training else 0.0
+        if drop == 0.0:
+            from turboprune_amd.ops.attention import sdpa
+            x = sdpa(q, k, v)  # torch SDPA, or fused HIP fwd when opted in
+        else:
+            x = torch.nn.functional.scaled_dot_product_attention(
+                q, k, v, dropout_p=drop)
         x = x.transpose(1, 2).reshape(B, N, C)
         return self.proj_drop(self.proj(x))
 
