@@ -1,0 +1,99 @@
+#!/bin/bash
+# Round-2 opening gpurun call: validate everything drafted off-device in
+# round 1, in one box session. Each step is independently timeout-capped
+# so one bad kernel cannot eat the call. Run as:
+#   /usr/local/graft/bin/gpurun --timeout 1500 -- 'bash scripts/round2_device_checks.sh'
+# then read gpurun_out/r2_checks/*.log and promote what passed:
+#   gemm256 PASS  -> wire TURBOPRUNE_GEMM256 into masked_linear dispatch
+#   attn PASS     -> flip TURBOPRUNE_ATTN=native for DeiT bench, A/B
+#   conv PASS     -> TURBOPRUNE_CONV=native bench A/B (the headline lever)
+#   multi-sgd     -> TURBOPRUNE_MULTI_SGD=1 bench A/B at small batch
+set -u
+OUT=gpurun_out/r2_checks
+mkdir -p "$OUT"
+
+run() {  # name timeout cmd...
+  local name=$1 tmo=$2; shift 2
+  echo "=== $name ==="
+  timeout "$tmo" "$@" > "$OUT/$name.log" 2>&1
+  echo "$name exit=$?" | tee -a "$OUT/summary.txt"
+  tail -3 "$OUT/$name.log"
+}
+
+run regression_gpu 600 python -m pytest tests -m gpu -x -q
+run gemm256 300 python scripts/validate_gemm256.py --bench
+run attn 300 python scripts/validate_attention.py --bench
+# native conv triple end-to-end through the autograd Function:
+run conv_native 300 python - <<'EOF'
+import json, torch, sys
+sys.path.insert(0, ".")
+import os
+os.environ["TURBOPRUNE_CONV"] = "native"
+from turboprune_amd.ops import conv_native
+ok = True
+for (cin, cout, k, s, hi) in [(64, 64, 3, 1, 56), (128, 128, 3, 2, 28),
+                              (256, 64, 1, 1, 56), (64, 256, 1, 2, 56)]:
+    torch.manual_seed(k)
+    x = (torch.rand(16, cin, hi, hi, device="cuda") - .5).bfloat16() \
+        .to(memory_format=torch.channels_last).requires_grad_()
+    w = ((torch.rand(cout, cin, k, k, device="cuda") - .5) * .1).bfloat16() \
+        .to(memory_format=torch.channels_last).requires_grad_()
+    y = conv_native.conv2d(x, w, None, s, k // 2,
+                           conv_native.NativeBackend)
+    gy = torch.randn_like(y)
+    gx, gw = torch.autograd.grad(y, [x, w], gy)
+    # reference grads via autograd on F.conv2d
+    x2 = x.detach().clone().requires_grad_(); w2 = w.detach().clone().requires_grad_()
+    y2 = torch.nn.functional.conv2d(x2, w2, None, s, k // 2)
+    y2.backward(gy)
+    ey = (y.float() - y2.float()).abs().max().item()
+    ex = (gx.float() - x2.grad.float()).abs().max().item()
+    ew = (gw.float() - w2.grad.float()).abs().max().item()
+    sc = max(y2.float().abs().max().item(), 1.0)
+    good = ey < .05 * sc and ex < .5 and ew < .5
+    ok &= good
+    print(json.dumps({"shape": [cin, cout, k, s, hi], "ey": round(ey, 4),
+                      "ex": round(ex, 4), "ew": round(ew, 4), "ok": good}))
+print("PASS" if ok else "FAIL")
+sys.exit(0 if ok else 1)
+EOF
+# multi-tensor SGD parity vs per-tensor on a real model step:
+run multi_sgd 300 python - <<'EOF'
+import os, sys, torch
+sys.path.insert(0, ".")
+from turboprune_amd.config import compose
+from turboprune_amd.models import build_model
+from turboprune_amd.optim import FusedMaskedSGD
+from turboprune_amd.ops import functional as TF
+
+def one(env):
+    os.environ.pop("TURBOPRUNE_MULTI_SGD", None)
+    if env:
+        os.environ["TURBOPRUNE_MULTI_SGD"] = "1"
+    torch.manual_seed(0)
+    cfg = compose("bench_resnet50_imagenet")
+    pm = build_model(cfg).to("cuda").to(memory_format=torch.channels_last)
+    pm.enable_caches(torch.bfloat16)
+    opt = FusedMaskedSGD(pm.parameters(), lr=0.1, momentum=0.9,
+                         weight_decay=1e-4, model=pm)
+    x = torch.randn(8, 3, 224, 224, device="cuda") \
+        .to(memory_format=torch.channels_last)
+    y = torch.randint(0, 1000, (8,), device="cuda")
+    for _ in range(3):
+        opt.zero_grad(set_to_none=True)
+        with torch.autocast("cuda", torch.bfloat16):
+            loss = TF.cross_entropy(pm(x), y)
+        loss.backward()
+        opt.step()
+    torch.cuda.synchronize()
+    return {n: p.detach().clone() for n, p in pm.named_parameters()}
+
+a = one(False)
+b = one(True)
+worst = max((a[n] - b[n]).abs().max().item() for n in a)
+print({"multi_vs_single_max_diff": worst})
+assert worst < 1e-5, worst
+print("PASS")
+EOF
+echo "---- summary ----"
+cat "$OUT/summary.txt"
