@@ -45,6 +45,37 @@ def parse_args():
     return p.parse_args()
 
 
+def build_result(args, world, elapsed, loss_val, bf16):
+    """The driver-contract JSON line (tests/test_bench_contract.py pins
+    the schema): whole-job aggregate value, max-over-ranks elapsed."""
+    images = args.steps * args.global_batch
+    ips = images / elapsed
+    epoch_minutes = 1_281_167 / ips / 60.0
+    return {
+        "metric": "imagenet_images_per_sec",
+        "value": round(ips, 1),
+        "unit": "images/s",
+        "n_gpus": world,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(1000.0 * elapsed / args.steps, 3),
+        "higher_is_better": True,
+        "scaling": "strong",
+        "vs_baseline": None,  # BASELINE.json "published" is empty
+        "dtype": "bf16" if bf16 else "fp32",
+        "data": "synthetic",
+        "config": {
+            "model": args.model,
+            "global_batch": args.global_batch,
+            "seq_len": 224,
+            "parallelism": f"dp{world}",
+            "sparsity": args.sparsity,
+            "epoch_time_min": round(epoch_minutes, 3),
+            "final_loss": round(loss_val, 4),
+        },
+    }
+
+
 def main():
     args = parse_args()
     rank = int(os.environ.get("RANK", 0))
@@ -171,32 +202,7 @@ def main():
         elapsed = float(t.item())
 
     if rank == 0:
-        images = args.steps * args.global_batch
-        ips = images / elapsed
-        epoch_minutes = 1_281_167 / ips / 60.0
-        result = {
-            "metric": "imagenet_images_per_sec",
-            "value": round(ips, 1),
-            "unit": "images/s",
-            "n_gpus": world,
-            "steps": args.steps,
-            "warmup": args.warmup,
-            "ms_per_step": round(1000.0 * elapsed / args.steps, 3),
-            "higher_is_better": True,
-            "scaling": "strong",
-            "vs_baseline": None,
-            "dtype": "bf16" if bf16 else "fp32",
-            "data": "synthetic",
-            "config": {
-                "model": args.model,
-                "global_batch": args.global_batch,
-                "seq_len": 224,
-                "parallelism": f"dp{world}",
-                "sparsity": args.sparsity,
-                "epoch_time_min": round(epoch_minutes, 3),
-                "final_loss": round(float(loss.item()), 4),
-            },
-        }
+        result = build_result(args, world, elapsed, float(loss.item()), bf16)
         print(json.dumps(result))
 
     if distributed:
