@@ -11,10 +11,14 @@ MI355X-specific choices:
   every forward (C3, weights-sized payload per step); here masks/weights
   are broadcast explicitly once per level / prune event
   (``broadcast_model_state``), identical observable behavior.
-- DDP bucket size defaults to 128 MB: xGMI is 7 point-to-point links per
-  GPU (~153 GB/s each), ring all-reduce is per-link bound, so fewer,
-  larger buckets amortize launch/latency better than the NVSwitch-tuned
-  25 MB default. Override with cfg ``experiment_params.bucket_cap_mb``.
+- DDP bucket size defaults to 32 MB. Two opposing pressures: xGMI is 7
+  point-to-point links per GPU (~153 GB/s each) and ring all-reduce is
+  per-link bound, favoring large messages — but ResNet50's gradients
+  total only ~102 MB, so a 128 MB cap degenerates to ONE bucket and the
+  all-reduce runs fully exposed after backward. 32 MB gives ~4 buckets
+  (4 MB per ring step per link — still link-efficient) and lets the
+  collectives overlap backward. Override with cfg
+  ``experiment_params.bucket_cap_mb``.
 - ``gradient_as_bucket_view=True``: no grad copy into buckets.
 """
 
@@ -29,7 +33,7 @@ import torch
 import torch.distributed as dist
 from torch.nn.parallel import DistributedDataParallel as DDP
 
-DEFAULT_BUCKET_CAP_MB = 128
+DEFAULT_BUCKET_CAP_MB = 32
 
 
 def world_info() -> Tuple[int, int, int]:
