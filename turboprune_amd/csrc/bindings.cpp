@@ -32,6 +32,8 @@ void sgd_step_multi_(std::vector<at::Tensor>, std::vector<at::Tensor>,
                      std::vector<at::Tensor>, double, double, double);
 at::Tensor attn_fwd(const at::Tensor&, const at::Tensor&,
                     const at::Tensor&, double);
+at::Tensor conv2d_implicit_gradin(const at::Tensor&, const at::Tensor&,
+                                  int64_t, int64_t, int64_t, int64_t);
 // fused NHWC batchnorm (batchnorm.hip)
 at::Tensor transpose2d(const at::Tensor&);
 at::Tensor conv2d_implicit_fwd(const at::Tensor&, const at::Tensor&,
@@ -107,6 +109,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "implicit-GEMM NHWC conv forward (experimental)");
   m.def("conv2d_implicit_wrw", &turboprune::conv2d_implicit_wrw,
         "implicit-GEMM NHWC conv weight gradient (experimental)");
+  m.def("conv2d_implicit_gradin", &turboprune::conv2d_implicit_gradin,
+        "implicit-GEMM NHWC conv input gradient, any stride, fused "
+        "dilated gather (experimental)");
   m.def("random_resized_crop", &turboprune::random_resized_crop,
         "fused bilinear RandomResizedCrop + flip + normalize");
   m.def("crop_translate", &turboprune::crop_translate,

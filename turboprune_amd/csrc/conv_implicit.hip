@@ -39,16 +39,24 @@ TP_DEVICE int lds_byte(int row, int k) {
 }
 }  // namespace conv_ig
 
-template <typename OutT, bool HAS_BIAS>
+// DIL: zero-insertion factor of the LOGICAL input image (fractional
+// stride). DIL=1 is the plain forward. For DIL>1 the input bounds
+// (Hi, Wi) are the DILATED dims; (Hc, Wc) are the compact tensor's dims
+// and only coordinates divisible by DIL hold data — everything else
+// gathers the zero page. This runs grad_input of a stride-DIL conv as
+// one fused forward conv of the output gradient with the
+// rotated-transposed weight (ops/conv_backward.py math), with no
+// materialized dilated tensor.
+template <typename OutT, bool HAS_BIAS, int DIL = 1>
 __launch_bounds__(256)
 __global__ void conv3x3_fwd_kernel(
-    const __hip_bfloat16* __restrict__ x,   // NHWC (N,Hi,Wi,Cin)
+    const __hip_bfloat16* __restrict__ x,   // NHWC (N,Hc,Wc,Cin) compact
     const __hip_bfloat16* __restrict__ wt,  // (Cout, KH*KW*Cin) tap-major
     OutT* __restrict__ y,                   // (M, Cout) row-major
     const float* __restrict__ bias,
     const __hip_bfloat16* __restrict__ zero_page, int Nb, int Hi, int Wi,
     int Cin, int Cout, int Ho, int Wo, int KH, int KW, int stride, int pad,
-    int grid_n) {
+    int grid_n, int Hc, int Wc) {
   using namespace conv_ig;
   __shared__ char smem[2 * 2 * BM * BK * 2];
   const int kTileBytes = BM * BK * 2;
@@ -93,10 +101,19 @@ __global__ void conv3x3_fwd_kernel(
         int n = (int)(r2 / Ho);
         int hi = ho * stride - pad + dh;
         int wi = wo * stride - pad + dw;
-        if (hi >= 0 && hi < Hi && wi >= 0 && wi < Wi) {
+        bool ok = hi >= 0 && hi < Hi && wi >= 0 && wi < Wi;
+        if (DIL > 1 && ok) {
+          if (hi % DIL || wi % DIL) {
+            ok = false;  // zero-inserted coordinate
+          } else {
+            hi /= DIL;
+            wi /= DIL;
+          }
+        }
+        if (ok) {
           int cin = cin0_base + src_blk * 8;
           srcA = reinterpret_cast<const char*>(
-                     x + (((int64_t)n * Hi + hi) * Wi + wi) * Cin + cin);
+                     x + (((int64_t)n * Hc + hi) * Wc + wi) * Cin + cin);
         } else {
           srcA = reinterpret_cast<const char*>(zero_page);
         }
@@ -235,13 +252,75 @@ at::Tensor conv2d_implicit_fwd(const at::Tensor& x_in,
     hipLaunchKernelGGL((conv3x3_fwd_kernel<__hip_bfloat16, true>), grid,
                        dim3(256), 0, stream, xp, wp, yp,
                        biasp.data_ptr<float>(), zp, Nb, Hi, Wi, Cin, Cout,
-                       Ho, Wo, KH, KW, (int)stride, (int)pad, grid_n);
+                       Ho, Wo, KH, KW, (int)stride, (int)pad, grid_n, Hi,
+                       Wi);
   else
     hipLaunchKernelGGL((conv3x3_fwd_kernel<__hip_bfloat16, false>), grid,
                        dim3(256), 0, stream, xp, wp, yp, nullptr, zp, Nb,
                        Hi, Wi, Cin, Cout, Ho, Wo, KH, KW, (int)stride,
-                       (int)pad, grid_n);
+                       (int)pad, grid_n, Hi, Wi);
   return y;
+}
+
+// grad_input of conv(x, w, stride, pad) as ONE fused kernel:
+//   gx = conv( dilate_stride(gy), rot180(w)^T, 1, k-1-pad )
+// with the zero-inserted gy coordinates resolved inside the im2col
+// gather (no materialized dilated tensor; ops/conv_backward.py is the
+// CPU-tested math oracle). gy: (N, Cout, Ho, Wo) channels_last bf16;
+// w: the ORIGINAL (Cout, Cin, KH, KW) channels_last weight.
+// Requires Cout % 64 == 0 (gy's channels are the contraction).
+at::Tensor conv2d_implicit_gradin(const at::Tensor& gy_in,
+                                  const at::Tensor& weight, int64_t Hi_l,
+                                  int64_t Wi_l, int64_t stride,
+                                  int64_t pad) {
+  TORCH_CHECK(gy_in.is_cuda() && gy_in.scalar_type() == at::kBFloat16);
+  auto gy = gy_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto w = weight.contiguous(at::MemoryFormat::ChannelsLast);
+  int Nb = gy.size(0), Cout = gy.size(1), Ho = gy.size(2), Wo = gy.size(3);
+  int Cin = w.size(1), KH = w.size(2), KW = w.size(3);
+  TORCH_CHECK(Cout % 64 == 0, "conv_gradin: Cout % 64 != 0");
+  TORCH_CHECK(KH == KW && stride >= 1);
+  // rotated-transposed weight, channels_last => (Cin, KH*KW*Cout) tap-major
+  auto w_rt = at::flip(w, {2, 3}).permute({1, 0, 2, 3})
+                  .contiguous(at::MemoryFormat::ChannelsLast);
+  int Hi = (int)Hi_l, Wi = (int)Wi_l;
+  int Hd = Hi + 2 * (int)pad - KH + 1;  // dilated image dims
+  int Wd = Wi + 2 * (int)pad - KW + 1;
+  TORCH_CHECK((Ho - 1) * stride + 1 <= Hd && (Wo - 1) * stride + 1 <= Wd,
+              "conv_gradin: inconsistent geometry");
+  int64_t M = (int64_t)Nb * Hi * Wi;
+
+  auto gx = at::empty({Nb, Cin, Hi, Wi},
+                      gy.options().memory_format(
+                          at::MemoryFormat::ChannelsLast));
+  static at::Tensor zero_page2;
+  if (!zero_page2.defined() || zero_page2.device() != gy.device())
+    zero_page2 = at::zeros({64}, gy.options());
+
+  int grid_m = (int)((M + conv_ig::BM - 1) / conv_ig::BM);
+  int grid_n = (Cin + conv_ig::BN - 1) / conv_ig::BN;
+  dim3 grid(grid_m * grid_n);
+  auto stream = at::hip::getCurrentHIPStream();
+  auto* gp = reinterpret_cast<const __hip_bfloat16*>(gy.data_ptr());
+  auto* wp = reinterpret_cast<const __hip_bfloat16*>(w_rt.data_ptr());
+  auto* xp = reinterpret_cast<__hip_bfloat16*>(gx.data_ptr());
+  auto* zp = reinterpret_cast<const __hip_bfloat16*>(zero_page2.data_ptr());
+  int new_pad = KH - 1 - (int)pad;
+  // roles: "x" = gy (dilated bounds Hd,Wd; compact Ho,Wo), "Cin" = Cout,
+  // "Cout" = Cin, output pixels = Hi x Wi, stride 1.
+  if (stride == 1)
+    hipLaunchKernelGGL((conv3x3_fwd_kernel<__hip_bfloat16, false, 1>),
+                       grid, dim3(256), 0, stream, gp, wp, xp, nullptr, zp,
+                       Nb, Hd, Wd, Cout, Cin, Hi, Wi, KH, KW, 1, new_pad,
+                       grid_n, Ho, Wo);
+  else if (stride == 2)
+    hipLaunchKernelGGL((conv3x3_fwd_kernel<__hip_bfloat16, false, 2>),
+                       grid, dim3(256), 0, stream, gp, wp, xp, nullptr, zp,
+                       Nb, Hd, Wd, Cout, Cin, Hi, Wi, KH, KW, 1, new_pad,
+                       grid_n, Ho, Wo);
+  else
+    TORCH_CHECK(false, "conv_gradin: stride > 2 unsupported");
+  return gx;
 }
 
 }  // namespace turboprune

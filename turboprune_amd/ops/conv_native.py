@@ -34,6 +34,12 @@ class TorchBackend:
         return F.conv2d(x, w, bias, stride, pad)
 
     @staticmethod
+    def gradin(gy, w, in_hw, stride, pad):
+        return conv_grad_input(gy, w, in_hw, stride, pad,
+                               conv_fn=lambda a, b, s, p:
+                               F.conv2d(a, b, None, s, p))
+
+    @staticmethod
     def wrw(gy, x, w_shape, stride, pad):
         return torch.nn.grad.conv2d_weight(x, w_shape, gy, stride, pad)
 
@@ -45,6 +51,14 @@ class NativeBackend:
     def fwd(x, w, bias, stride, pad):
         from turboprune_amd.ops._backend import extension
         return extension().conv2d_implicit_fwd(x, w, bias, stride, pad)
+
+    @staticmethod
+    def gradin(gy, w, in_hw, stride, pad):
+        # one fused kernel: rotated-weight forward conv with the
+        # zero-inserted gy coordinates resolved inside the im2col gather
+        from turboprune_amd.ops._backend import extension
+        return extension().conv2d_implicit_gradin(
+            gy, w, in_hw[0], in_hw[1], stride, pad)
 
     @staticmethod
     def wrw(gy, x, w_shape, stride, pad):
@@ -67,9 +81,8 @@ class ConvImplicitFn(torch.autograd.Function):
         stride, pad, backend = ctx.stride, ctx.pad, ctx.backend
         gx = gw = gb = None
         if ctx.needs_input_grad[0]:
-            gx = conv_grad_input(
-                gy, w, (x.shape[2], x.shape[3]), stride, pad,
-                conv_fn=lambda a, b, s, p: backend.fwd(a, b, None, s, p))
+            gx = backend.gradin(gy, w, (x.shape[2], x.shape[3]), stride,
+                                pad)
         if ctx.needs_input_grad[1]:
             gw = backend.wrw(gy, x, w.shape, stride, pad)
             if gw.dtype != w.dtype:
