@@ -184,3 +184,37 @@ def test_sharded_reader_rrc_path(tmp_path):
                          False, image_size=32)
     xv, yv = next(iter(va))
     assert xv.shape == (16, 3, 32, 32)
+
+
+def test_sharded_equal_steps_across_ranks(tmp_path):
+    """Unequal shard counts/sizes must NOT give ranks different step
+    counts (per-step DDP all-reduce would deadlock): every rank stops at
+    the global minimum batch budget, derived deterministically from all
+    shard headers."""
+    import torch
+
+    from turboprune_amd.data.imagenet import ShardedImageNet
+    root = tmp_path / "inet"
+    (root / "train").mkdir(parents=True)
+    # 3 shards of sizes 10, 10, 6 -> world 2: rank0 gets [10, 6]=4
+    # batches@bs4, rank1 gets [10]=2 batches -> both must run 2
+    for i, n in enumerate([10, 10, 6]):
+        torch.save({"images": torch.randint(0, 255, (n, 3, 8, 8),
+                                            dtype=torch.uint8),
+                    "labels": torch.randint(0, 10, (n,))},
+                   root / "train" / f"shard_{i:03d}.pt")
+    counts = []
+    for rank in range(2):
+        ld = ShardedImageNet(str(root), "train", batch_size=4,
+                             device=torch.device("cpu"), train=True,
+                             world_size=2, rank=rank, image_size=8)
+        counts.append(sum(1 for _ in ld))
+        assert len(ld) == counts[-1]
+    assert counts[0] == counts[1] == 2
+
+    # more ranks than shards: everyone gets one shard, equal budget
+    for rank in range(4):
+        ld = ShardedImageNet(str(root), "train", batch_size=4,
+                             device=torch.device("cpu"), train=True,
+                             world_size=4, rank=rank, image_size=8)
+        assert sum(1 for _ in ld) == 1  # min shard 6 // 4 = 1

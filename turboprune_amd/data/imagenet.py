@@ -98,12 +98,26 @@ class ShardedImageNet:
                  image_size: int = 224):
         self.image_size = image_size
         self.dir = os.path.join(root, split)
-        self.paths: List[str] = sorted(
+        all_paths: List[str] = sorted(
             os.path.join(self.dir, f) for f in os.listdir(self.dir)
             if f.startswith("shard_") and f.endswith(".pt"))
-        if not self.paths:
+        if not all_paths:
             raise FileNotFoundError(f"no shards under {self.dir}")
-        self.paths = self.paths[rank::world_size] or self.paths[:1]
+        # Every rank must run the SAME number of steps per epoch or the
+        # DDP all-reduce deadlocks (ranks with more shards/bigger shards
+        # would keep stepping). Each rank reads every shard's header
+        # (mmap: metadata only), derives every rank's batch budget, and
+        # stops at the global minimum — deterministic, no communication.
+        sizes = [self._shard_len(p) for p in all_paths]
+        if len(all_paths) < world_size:
+            self.paths = [all_paths[rank % len(all_paths)]]
+            self.steps_per_epoch = min(sizes) // batch_size
+        else:
+            self.paths = all_paths[rank::world_size]
+            budgets = [sum(s // batch_size
+                           for s in sizes[r::world_size])
+                       for r in range(world_size)]
+            self.steps_per_epoch = min(budgets)
         self.batch_size = batch_size
         self.device = device
         self.train = train
@@ -115,12 +129,19 @@ class ShardedImageNet:
         self._copy_stream = (torch.cuda.Stream(device)
                              if device.type == "cuda" else None)
 
+    @staticmethod
+    def _shard_len(path: str) -> int:
+        blob = torch.load(path, map_location="cpu", weights_only=True,
+                          mmap=True)
+        return int(blob["images"].shape[0])
+
     def _load_shard(self, path: str):
         blob = torch.load(path, map_location="cpu", weights_only=True,
                           mmap=True)
         return blob["images"], blob["labels"]
 
     def __iter__(self):
+        emitted = 0
         order = list(range(len(self.paths)))
         if self.train:
             g = torch.Generator().manual_seed(self.seed + self.epoch)
@@ -165,14 +186,14 @@ class ShardedImageNet:
                     x = augment.normalize_u8(raw, self._mean, self._std,
                                              self.dtype, flip)
                 yield x, labels[idx].to(self.device, non_blocking=True)
+                emitted += 1
+                if emitted >= self.steps_per_epoch:
+                    self.epoch += 1
+                    return
         self.epoch += 1
 
     def __len__(self) -> int:
-        # approximate: full shards assumed equal-sized
-        first = torch.load(self.paths[0], map_location="cpu",
-                           weights_only=True, mmap=True)
-        per_shard = first["images"].shape[0] // self.batch_size
-        return per_shard * len(self.paths)
+        return self.steps_per_epoch
 
 
 class ImageNetLoaders:
