@@ -62,6 +62,18 @@ def main():
         ok &= good
         print(json.dumps({"refcheck": [M, N, K], "max_err": round(err, 5),
                           "tol": round(tol, 5), "ok": good}))
+    # A = I with asymmetric B: catches a transposed / misrouted C that
+    # random-operand refchecks can miss (guide §5.4)
+    I = torch.eye(256, device="cuda").bfloat16()
+    Ba = torch.zeros(256, 256, device="cuda")
+    Ba[3, 7] = 1.0
+    Ba[200, 40] = -2.0
+    got = ext.gemm_bt_256(I, Ba.bfloat16(), None, False).float()
+    good = (got[7, 3].item() == 1.0 and got[40, 200].item() == -2.0
+            and got.abs().sum().item() == 3.0)
+    ok &= good
+    print(json.dumps({"refcheck": "identity-asymB", "ok": good}))
+
     # bias
     torch.manual_seed(1)
     A = (torch.rand(256, 512, device="cuda") - 0.5).bfloat16()

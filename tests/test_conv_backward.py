@@ -40,3 +40,26 @@ def test_dilate_geometry_stride2():
     assert d.shape == (2, 3, 8, 8)
     assert torch.equal(d[:, :, ::2, ::2][:, :, :4, :4], gy)
     assert d[:, :, 1::2].abs().sum() == 0
+
+
+def test_grad_input_fuzz_geometries():
+    """Randomized geometries beyond ResNet50 (k up to 5, stride up to 3,
+    arbitrary pad) against autograd."""
+    import random
+    rng = random.Random(7)
+    for _ in range(25):
+        k = rng.choice([1, 3, 5])
+        s = rng.choice([1, 2, 3])
+        pad = rng.randint(0, k - 1) if k > 1 else 0
+        cin, cout = rng.choice([4, 8]), rng.choice([4, 8])
+        hi = rng.randint(k + s, 20)
+        torch.manual_seed(hi)
+        x = torch.randn(2, cin, hi, hi, requires_grad=True)
+        w = torch.randn(cout, cin, k, k)
+        y = F.conv2d(x, w, None, s, pad)
+        if y.shape[-1] < 1:
+            continue
+        gy = torch.randn_like(y)
+        y.backward(gy)
+        gx = conv_grad_input(gy, w, (hi, hi), s, pad)
+        torch.testing.assert_close(gx, x.grad, rtol=1e-4, atol=1e-4)
