@@ -195,3 +195,29 @@ def test_imagenet_synthetic_end_to_end(tmp_path):
     expt_dir = run(cfg)
     assert os.path.exists(os.path.join(expt_dir, "checkpoints",
                                        "model_level_0.pt"))
+
+
+def test_cyclic_driver_main(tmp_path, monkeypatch):
+    """run_cyclic_training_experiment.main end-to-end on CPU (the
+    reference's cyclic driver crashes as shipped)."""
+    import os
+
+    from run_cyclic_training_experiment import main as cyclic_main
+    monkeypatch.setenv("TURBOPRUNE_SYNTHETIC_CIFAR", "1")
+    expt_dir = cyclic_main([
+        "--config-name=cifar10_er_erk",
+        f"experiment_params.base_dir={tmp_path}",
+        "experiment_params.epochs_per_level=2",
+        "dataset_params.batch_size=32",
+        "+dataset_params.synthetic_size=64",
+        "cyclic_training.num_cycles=2",
+        "cyclic_training.strategy=constant",
+        "pruning_params.target_sparsity=0.5",
+    ])
+    assert os.path.isdir(expt_dir)
+    level0 = os.path.join(expt_dir, "metrics", "level_wise_metrics",
+                          "level_0_metrics.csv")
+    assert os.path.exists(level0)
+    with open(level0) as f:
+        header = f.readline()
+    assert "cycle" in header
