@@ -30,6 +30,14 @@ at::Tensor gemm_bt_256(const at::Tensor&, const at::Tensor&,
 void sgd_step_multi_(std::vector<at::Tensor>, std::vector<at::Tensor>,
                      std::vector<at::Tensor>, std::vector<at::Tensor>,
                      std::vector<at::Tensor>, double, double, double);
+std::vector<at::Tensor> sgd_multi_plan(std::vector<at::Tensor>,
+                                       std::vector<at::Tensor>,
+                                       std::vector<at::Tensor>,
+                                       std::vector<at::Tensor>,
+                                       std::vector<at::Tensor>);
+void sgd_step_multi_planned_(const at::Tensor&, const at::Tensor&, bool,
+                             bool, bool, bool, bool, double, double,
+                             double);
 at::Tensor attn_fwd(const at::Tensor&, const at::Tensor&,
                     const at::Tensor&, double);
 at::Tensor conv2d_implicit_gradin(const at::Tensor&, const at::Tensor&,
@@ -81,6 +89,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgd_step_multi_", &turboprune::sgd_step_multi_,
         "multi-tensor fused SGD step (one launch over many params; "
         "uniform flags per call — opt-in via TURBOPRUNE_MULTI_SGD=1)");
+  m.def("sgd_multi_plan", &turboprune::sgd_multi_plan,
+        "build the device descriptor table + block map for the "
+        "multi-tensor SGD step (cache keyed by operand data_ptrs)");
+  m.def("sgd_step_multi_planned_", &turboprune::sgd_step_multi_planned_,
+        "launch the multi-tensor SGD step from a prebuilt plan");
   m.def("attn_fwd", &turboprune::attn_fwd,
         "EXPERIMENTAL fused flash-style attention forward, head_dim 64 "
         "(unwired; opt-in TURBOPRUNE_ATTN=native after device validation)");

@@ -60,14 +60,18 @@ __global__ void sgd_multi_kernel(const int64_t* __restrict__ desc,
   }
 }
 
-void sgd_step_multi_(std::vector<at::Tensor> ws, std::vector<at::Tensor> gs,
-                     std::vector<at::Tensor> bufs,
-                     std::vector<at::Tensor> masks,
-                     std::vector<at::Tensor> caches, double lr,
-                     double momentum, double wd) {
+// Build the device-side descriptor table + block map once; pointers are
+// stable in steady state, so the Python side caches the plan keyed by
+// every operand's data_ptr (any reallocation changes the key and
+// triggers a rebuild — staleness is impossible by construction).
+std::vector<at::Tensor> sgd_multi_plan(std::vector<at::Tensor> ws,
+                                       std::vector<at::Tensor> gs,
+                                       std::vector<at::Tensor> bufs,
+                                       std::vector<at::Tensor> masks,
+                                       std::vector<at::Tensor> caches) {
   size_t n = ws.size();
-  TORCH_CHECK(n > 0 && gs.size() == n, "sgd_step_multi_: empty/mismatched");
-  bool has_m = momentum != 0.0 && bufs.size() == n;
+  TORCH_CHECK(n > 0 && gs.size() == n, "sgd_multi_plan: empty/mismatched");
+  bool has_m = bufs.size() == n;
   bool has_mask = masks.size() == n;
   bool has_cache = caches.size() == n;
   auto grad_t = gs[0].scalar_type();
@@ -118,12 +122,20 @@ void sgd_step_multi_(std::vector<at::Tensor> ws, std::vector<at::Tensor> gs,
       bm.push_back(c);
     }
   }
-  int blocks = (int)(bm.size() / 2);
   auto bm_cpu = at::from_blob(bm.data(), {(int64_t)bm.size()},
                               at::TensorOptions().dtype(at::kInt)).clone();
   auto dev = ws[0].device();
-  auto desc = desc_cpu.to(dev);
-  auto bmap = bm_cpu.to(dev);
+  return {desc_cpu.to(dev), bm_cpu.to(dev)};
+}
+
+void sgd_step_multi_planned_(const at::Tensor& desc, const at::Tensor& bmap,
+                             bool has_m, bool has_mask, bool has_cache,
+                             bool grad_bf16, bool cache_bf16, double lr,
+                             double momentum, double wd) {
+  TORCH_CHECK(desc.is_cuda() && desc.scalar_type() == at::kLong);
+  TORCH_CHECK(bmap.is_cuda() && bmap.scalar_type() == at::kInt);
+  int blocks = (int)(bmap.numel() / 2);
+  has_m = has_m && momentum != 0.0;
   auto stream = at::hip::getCurrentHIPStream();
 
 #define TP_MSGD(GT, CT, MOM, MASKF, CACHEF)                              \
@@ -144,18 +156,34 @@ void sgd_step_multi_(std::vector<at::Tensor> ws, std::vector<at::Tensor> gs,
     }                                                                    \
   } while (0)
 
-  if (grad_t == at::kFloat) {
-    if (cache_t == at::kBFloat16) TP_MSGD_FLAGS(float, __hip_bfloat16);
+  if (!grad_bf16) {
+    if (cache_bf16) TP_MSGD_FLAGS(float, __hip_bfloat16);
     else TP_MSGD_FLAGS(float, float);
-  } else if (grad_t == at::kBFloat16) {
-    if (cache_t == at::kBFloat16)
-      TP_MSGD_FLAGS(__hip_bfloat16, __hip_bfloat16);
-    else TP_MSGD_FLAGS(__hip_bfloat16, float);
   } else {
-    TORCH_CHECK(false, "sgd_step_multi_: unsupported grad dtype");
+    if (cache_bf16) TP_MSGD_FLAGS(__hip_bfloat16, __hip_bfloat16);
+    else TP_MSGD_FLAGS(__hip_bfloat16, float);
   }
 #undef TP_MSGD_FLAGS
 #undef TP_MSGD
+}
+
+// build + launch in one call (convenience / non-cached path)
+void sgd_step_multi_(std::vector<at::Tensor> ws, std::vector<at::Tensor> gs,
+                     std::vector<at::Tensor> bufs,
+                     std::vector<at::Tensor> masks,
+                     std::vector<at::Tensor> caches, double lr,
+                     double momentum, double wd) {
+  size_t n = ws.size();
+  bool has_m = momentum != 0.0 && bufs.size() == n;
+  bool has_mask = masks.size() == n;
+  bool has_cache = caches.size() == n;
+  bool grad_bf16 = gs[0].scalar_type() == at::kBFloat16;
+  bool cache_bf16 =
+      has_cache && caches[0].scalar_type() == at::kBFloat16;
+  auto plan = sgd_multi_plan(std::move(ws), std::move(gs), std::move(bufs),
+                             std::move(masks), std::move(caches));
+  sgd_step_multi_planned_(plan[0], plan[1], has_m, has_mask, has_cache,
+                          grad_bf16, cache_bf16, lr, momentum, wd);
 }
 
 }  // namespace turboprune

@@ -117,13 +117,32 @@ class FusedMaskedSGD(torch.optim.SGD):
                     p.add_(d_p, alpha=-lr)
                     if cache is not None and layer is not None:
                         layer.refresh_cache()
-            for (gdt, has_mask, has_cache, _cdt), items in buckets.items():
+            for (gdt, has_mask, has_cache, cdt), items in buckets.items():
                 ext = _backend.extension()
-                ext.sgd_step_multi_(
-                    [it[0] for it in items], [it[1] for it in items],
-                    [it[2] for it in items] if momentum != 0 else [],
-                    [it[3] for it in items] if has_mask else [],
-                    [it[4] for it in items] if has_cache else [],
+                ws = [it[0] for it in items]
+                gs = [it[1] for it in items]
+                bufs = [it[2] for it in items] if momentum != 0 else []
+                masks = [it[3] for it in items] if has_mask else []
+                caches = [it[4] for it in items] if has_cache else []
+                # plan cache keyed by every operand's data_ptr: pointers
+                # are stable in steady state (params/bufs/masks fixed,
+                # allocator reuses grad blocks); ANY reallocation changes
+                # the key and rebuilds — a stale plan cannot be hit.
+                key = tuple(t.data_ptr()
+                            for grp in (ws, gs, bufs, masks, caches)
+                            for t in grp)
+                plans = getattr(self, "_multi_plans", None)
+                if plans is None:
+                    plans = self._multi_plans = {}
+                plan = plans.get(key)
+                if plan is None:
+                    if len(plans) > 64:  # ptr churn: don't grow unbounded
+                        plans.clear()
+                    plan = plans[key] = ext.sgd_multi_plan(
+                        ws, gs, bufs, masks, caches)
+                ext.sgd_step_multi_planned_(
+                    plan[0], plan[1], momentum != 0, has_mask, has_cache,
+                    gdt == torch.bfloat16, cdt == torch.bfloat16,
                     float(lr), float(momentum), float(wd))
         return loss
 
