@@ -301,6 +301,32 @@ at::Tensor gemm_bf16(const at::Tensor& A, const at::Tensor& B, bool /*ta*/,
   return gemm_bt(A, B, c10::nullopt, /*out_fp32=*/false);
 }
 
+// Opt-in routing to the EXPERIMENTAL 256^2 8-phase kernel
+// (gemm_256_8phase.hip) — promotion path for round 2: flip
+// TURBOPRUNE_GEMM256=1 after scripts/validate_gemm256.py passes on
+// device. Large bf16-out shapes only; everything else stays on the
+// validated 128^2 kernel.
+at::Tensor gemm_bt_256(const at::Tensor&, const at::Tensor&,
+                       const c10::optional<at::Tensor>&, bool);
+
+static bool use_g256() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("TURBOPRUNE_GEMM256");
+    v = (e && e[0] == '1') ? 1 : 0;
+  }
+  return v == 1;
+}
+
+static at::Tensor gemm_route(const at::Tensor& A, const at::Tensor& B,
+                             const c10::optional<at::Tensor>& bias,
+                             bool out_fp32) {
+  if (use_g256() && !out_fp32 && A.size(0) >= 256 && B.size(0) >= 256 &&
+      A.size(1) >= 256)
+    return gemm_bt_256(A, B, bias, out_fp32);
+  return gemm_bt(A, B, bias, out_fp32);
+}
+
 bool masked_linear_available(const at::Tensor& x, const at::Tensor& w) {
   if (!x.is_cuda() || x.scalar_type() != at::kBFloat16 ||
       w.scalar_type() != at::kBFloat16)
@@ -320,7 +346,7 @@ at::Tensor linear_fwd(const at::Tensor& x, const at::Tensor& w,
   int64_t K = sizes.back();
   int64_t M = x.numel() / K;
   auto x2 = x.reshape({M, K}).contiguous();
-  auto y = gemm_bt(x2, w.contiguous(), bias, /*out_fp32=*/false);
+  auto y = gemm_route(x2, w.contiguous(), bias, /*out_fp32=*/false);
   sizes.back() = w.size(0);
   return y.reshape(sizes);
 }
@@ -338,10 +364,10 @@ std::tuple<at::Tensor, at::Tensor> linear_bwd(const at::Tensor& grad_y,
   // measured ~1.1 TB/s and dominated the DeiT step)
   // grad_x (M,K) = gy (M,N) @ w (N,K): B_t = w^T (K,N) K-major in N
   auto wT = transpose2d(w.contiguous());
-  auto gx = gemm_bt(gy2, wT, c10::nullopt, false);
+  auto gx = gemm_route(gy2, wT, c10::nullopt, false);
   // grad_w (N,K) = gy^T (N,M) @ x (M,K): A = gy^T, B_t = x^T (K,M)
-  auto gw = gemm_bt(transpose2d(gy2), transpose2d(x2), c10::nullopt,
-                    false);
+  auto gw = gemm_route(transpose2d(gy2), transpose2d(x2), c10::nullopt,
+                       false);
   auto x_sizes = x.sizes().vec();
   return {gx.reshape(x_sizes), gw};
 }
