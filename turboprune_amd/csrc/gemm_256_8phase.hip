@@ -204,8 +204,17 @@ __global__ void gemm256_kernel(const __hip_bfloat16* __restrict__ A,
       __builtin_amdgcn_s_setprio(0);
       // -- tile-boundary retire: all but the 2 newest half-tile
       //    stagings complete in this wave; the barrier makes that hold
-      //    across waves before the next tile's p0 ds_reads
-      if (p == 3) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+      //    across waves before the next tile's p0 ds_reads. TAIL: when
+      //    this tile staged no B half-tiles (t+2 past the end), the
+      //    newest 2 loads ARE the next tile's A halves — drain fully
+      //    (caught by tests/test_gemm256_schedule.py's adversarial
+      //    timing model).
+      if (p == 3) {
+        if (t + 2 < total_kt)
+          asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
       __builtin_amdgcn_s_barrier();
     }
   }
