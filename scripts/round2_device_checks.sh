@@ -95,5 +95,19 @@ print({"multi_vs_single_max_diff": worst})
 assert worst < 1e-5, worst
 print("PASS")
 EOF
+# ---- promotion A/B benches (short; compare value fields) -------------
+run bench_baseline 420 python bench.py --steps 15 --warmup 5
+run bench_conv_native 420 env TURBOPRUNE_CONV=native \
+    python bench.py --steps 15 --warmup 5
+run bench_multi_sgd 420 env TURBOPRUNE_MULTI_SGD=1 \
+    python bench.py --steps 15 --warmup 5
+run bench_deit_baseline 420 python bench.py --model deit_small \
+    --global-batch 256 --steps 15 --warmup 5
+run bench_deit_g256 420 env TURBOPRUNE_GEMM256=1 python bench.py \
+    --model deit_small --global-batch 256 --steps 15 --warmup 5
+run bench_deit_attn 420 env TURBOPRUNE_ATTN=native python bench.py \
+    --model deit_small --global-batch 256 --steps 15 --warmup 5
+
 echo "---- summary ----"
 cat "$OUT/summary.txt"
+grep -h '"value"' "$OUT"/bench_*.log 2>/dev/null | tail -12
