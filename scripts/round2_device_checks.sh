@@ -2,7 +2,9 @@
 # Round-2 opening gpurun call: validate everything drafted off-device in
 # round 1, in one box session. Each step is independently timeout-capped
 # so one bad kernel cannot eat the call. Run as:
-#   /usr/local/graft/bin/gpurun --timeout 1500 -- 'bash scripts/round2_device_checks.sh'
+#   /usr/local/graft/bin/gpurun --timeout 2400 -- 'bash scripts/round2_device_checks.sh'
+# (realistic runtime ~20-30 min: the per-step timeouts are caps, not
+#  expected durations; most cost is per-invocation torch/model init)
 # then read gpurun_out/r2_checks/*.log and promote what passed:
 #   gemm256 PASS  -> wire TURBOPRUNE_GEMM256 into masked_linear dispatch
 #   attn PASS     -> flip TURBOPRUNE_ATTN=native for DeiT bench, A/B
