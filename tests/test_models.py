@@ -93,3 +93,28 @@ def test_build_model_unknown_raises():
     cfg = compose("cifar10_er_erk", ["model_params.model_name=nonexistent"])
     with pytest.raises(ValueError, match="unknown model"):
         build_model(cfg)
+
+
+def test_resnet152_and_wide_variants():
+    """Extended zoo: torchvision-named deep/wide ResNets (torchvision is
+    not installed in the image; the native zoo is the model surface)."""
+    import torch
+
+    from turboprune_amd.models import available_models, build_model
+    from turboprune_amd.config import compose
+    names = available_models()
+    assert {"resnet152", "wide_resnet50_2", "wide_resnet101_2"} <= set(names)
+
+    cfg = compose("bench_resnet50_imagenet",
+                  ["model_params.model_name=wide_resnet50_2"])
+    pm = build_model(cfg, num_classes=10)
+    x = torch.randn(2, 3, 64, 64)
+    y = pm(x)
+    assert y.shape == (2, 10)
+    # wide: bottleneck conv2 of layer1 has 128 channels (64 * 128/64)
+    w = pm.model.layer1[0].conv2.weight
+    assert w.shape[0] == 128 and w.shape[1] == 128
+    # resnet50 unchanged by the width plumbing
+    cfg50 = compose("bench_resnet50_imagenet")
+    pm50 = build_model(cfg50, num_classes=10)
+    assert pm50.model.layer1[0].conv2.weight.shape[0] == 64

@@ -68,13 +68,15 @@ class Bottleneck(nn.Module):
     expansion = 4
 
     def __init__(self, inplanes: int, planes: int, stride: int = 1,
-                 downsample: Optional[nn.Module] = None):
+                 downsample: Optional[nn.Module] = None,
+                 base_width: int = 64):
         super().__init__()
-        self.conv1 = conv1x1(inplanes, planes)
-        self.bn1 = FusedBatchNorm2d(planes)
-        self.conv2 = conv3x3(planes, planes, stride)
-        self.bn2 = FusedBatchNorm2d(planes)
-        self.conv3 = conv1x1(planes, planes * self.expansion)
+        width = int(planes * (base_width / 64.0))
+        self.conv1 = conv1x1(inplanes, width)
+        self.bn1 = FusedBatchNorm2d(width)
+        self.conv2 = conv3x3(width, width, stride)
+        self.bn2 = FusedBatchNorm2d(width)
+        self.conv3 = conv1x1(width, planes * self.expansion)
         self.bn3 = FusedBatchNorm2d(planes * self.expansion)
         self.relu = nn.ReLU(inplace=True)
         self.downsample = downsample
@@ -93,9 +95,10 @@ class Bottleneck(nn.Module):
 class ResNet(nn.Module):
     def __init__(self, block: Type[Union[BasicBlock, Bottleneck]],
                  layers: List[int], num_classes: int = 1000,
-                 cifar_stem: bool = False):
+                 cifar_stem: bool = False, width_per_group: int = 64):
         super().__init__()
         self.inplanes = 64
+        self.base_width = width_per_group
         if cifar_stem:
             # CIFAR surgery: 3x3 stride-1 stem, no maxpool
             # (reference: custom_models.py:197-215)
@@ -133,9 +136,11 @@ class ResNet(nn.Module):
                 conv1x1(self.inplanes, planes * block.expansion, stride),
                 FusedBatchNorm2d(planes * block.expansion),
             )
-        layers = [block(self.inplanes, planes, stride, downsample)]
+        kw = {} if block is BasicBlock else {"base_width": self.base_width}
+        layers = [block(self.inplanes, planes, stride, downsample, **kw)]
         self.inplanes = planes * block.expansion
-        layers += [block(self.inplanes, planes) for _ in range(1, blocks)]
+        layers += [block(self.inplanes, planes, **kw)
+                   for _ in range(1, blocks)]
         return nn.Sequential(*layers)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -163,3 +168,19 @@ def resnet50(num_classes: int = 1000, cifar_stem: bool = False) -> ResNet:
 
 def resnet101(num_classes: int = 1000, cifar_stem: bool = False) -> ResNet:
     return ResNet(Bottleneck, [3, 4, 23, 3], num_classes, cifar_stem)
+
+
+def resnet152(num_classes: int = 1000, cifar_stem: bool = False) -> ResNet:
+    return ResNet(Bottleneck, [3, 8, 36, 3], num_classes, cifar_stem)
+
+
+def wide_resnet50_2(num_classes: int = 1000,
+                    cifar_stem: bool = False) -> ResNet:
+    return ResNet(Bottleneck, [3, 4, 6, 3], num_classes, cifar_stem,
+                  width_per_group=128)
+
+
+def wide_resnet101_2(num_classes: int = 1000,
+                     cifar_stem: bool = False) -> ResNet:
+    return ResNet(Bottleneck, [3, 4, 23, 3], num_classes, cifar_stem,
+                  width_per_group=128)

@@ -30,7 +30,8 @@ _FACTORIES = {}
 for _mod in (resnet_models, vgg_models, deit_models):
     for _name in dir(_mod):
         _fn = getattr(_mod, _name)
-        if callable(_fn) and (_name.startswith(("resnet", "vgg", "local_deit"))):
+        if callable(_fn) and (_name.startswith(
+                ("resnet", "wide_resnet", "vgg", "local_deit"))):
             _FACTORIES[_name] = _fn
 
 
