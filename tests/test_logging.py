@@ -62,3 +62,13 @@ def test_wandb_shim_noop():
     run = WandbShim(enabled=False, project=None, name="x", config={})
     run.log({"a": 1})  # must not raise
     run.finish()
+
+
+def test_display_training_info_smoke(capsys):
+    from turboprune_amd.config import compose
+    from turboprune_amd.utils.console import display_training_info
+    cfg = compose("cifar10_er_erk")
+    display_training_info({"dataset": "CIFAR10", "model": "resnet18"},
+                          cfg.to_dict())
+    out = capsys.readouterr().out
+    assert "CIFAR10" in out or "resnet18" in out or len(out) > 0
