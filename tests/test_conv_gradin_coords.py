@@ -63,3 +63,40 @@ def test_kernel_coordinate_math_matches_autograd(k, stride, pad, hi):
     y.backward(gy)
     gx = gradin_via_kernel_coords(gy, w, hi, hi, stride, pad)
     torch.testing.assert_close(gx, x.grad, rtol=1e-4, atol=1e-4)
+
+
+def conv_fwd_via_kernel_coords(x, w, stride, pad):
+    """Mirror of conv3x3_fwd_kernel's im2col gather (DIL=1): source =
+    x[n, ho*stride-pad+dh, wo*stride-pad+dw] with zero-page fallback;
+    B = channels_last weight viewed (Cout, KH*KW*Cin) tap-major."""
+    N, Cin, Hi, Wi = x.shape
+    Cout, _, KH, KW = w.shape
+    Ho = (Hi + 2 * pad - KH) // stride + 1
+    Wo = (Wi + 2 * pad - KW) // stride + 1
+    B = w.permute(0, 2, 3, 1).reshape(Cout, KH * KW * Cin)
+    y = torch.zeros(N, Cout, Ho, Wo)
+    for n in range(N):
+        for ho in range(Ho):
+            for wo in range(Wo):
+                a = torch.zeros(KH * KW * Cin)
+                for dh in range(KH):
+                    for dw in range(KW):
+                        hi = ho * stride - pad + dh
+                        wi = wo * stride - pad + dw
+                        if 0 <= hi < Hi and 0 <= wi < Wi:
+                            tap = dh * KW + dw
+                            a[tap * Cin:(tap + 1) * Cin] = x[n, :, hi, wi]
+                y[n, :, ho, wo] = B @ a
+    return y
+
+
+@pytest.mark.parametrize("k,stride,pad", [(3, 1, 1), (3, 2, 1),
+                                          (1, 1, 0), (1, 2, 0),
+                                          (7, 2, 3)])
+def test_fwd_kernel_coordinate_math(k, stride, pad):
+    torch.manual_seed(1)
+    x = torch.randn(2, 3, 12, 12)
+    w = torch.randn(4, 3, k, k)
+    got = conv_fwd_via_kernel_coords(x, w, stride, pad)
+    ref = F.conv2d(x, w, None, stride, pad)
+    torch.testing.assert_close(got, ref, rtol=1e-4, atol=1e-4)

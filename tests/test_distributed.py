@@ -214,3 +214,63 @@ def test_run_experiment_distributed_two_ranks():
     # ER-ERK clamping makes realized sparsity exceed the request
     # (reference formula, see test_pruning) — just require a real prune
     assert 0.4 < zeros / total < 0.75
+
+
+def _worker_sharded(rank, world, port, q):
+    """DDP epoch over the SHARDED loader with unequal shard counts: the
+    deterministic equal-step budget must keep both ranks in lockstep
+    (without it the per-step all-reduce deadlocks)."""
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    try:
+        import tempfile
+
+        from run_experiment import run
+        from turboprune_amd.config import compose
+
+        shards = os.environ["TP_TEST_SHARDS"]
+        tmp = tempfile.mkdtemp(prefix=f"tp_shard_{rank}_")
+        cfg = compose("bench_resnet50_imagenet", [
+            "model_params.model_name=resnet18",
+            "experiment_params.epochs_per_level=1",
+            "experiment_params.distributed=true",
+            "dataset_params.total_batch_size=8",
+            "dataset_params.dataloader_type=native",
+            f"dataset_params.data_root_dir={shards}",
+            f"experiment_params.base_dir={tmp}/experiments",
+            "pruning_params=pai_er_erk",
+            "pruning_params.target_sparsity=0.5",
+        ])
+        expt_dir = run(cfg)
+        q.put(("ok", rank, expt_dir if rank == 0 else ""))
+    except BaseException as e:  # noqa: BLE001
+        q.put(("err", rank, repr(e)))
+        raise
+    finally:
+        for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR",
+                  "MASTER_PORT"):
+            os.environ.pop(k, None)
+
+
+@pytest.mark.timeout(600)
+def test_run_experiment_sharded_loader_two_ranks(tmp_path, monkeypatch):
+    root = tmp_path / "shards"
+    g = torch.Generator().manual_seed(3)
+    for split, sizes in (("train", [10, 10, 6]), ("val", [8])):
+        (root / split).mkdir(parents=True)
+        for i, n in enumerate(sizes):
+            torch.save(
+                {"images": torch.randint(0, 255, (n, 3, 64, 64),
+                                         dtype=torch.uint8, generator=g),
+                 "labels": torch.randint(0, 1000, (n,), generator=g)},
+                root / split / f"shard_{i:03d}.pt")
+    monkeypatch.setenv("TP_TEST_SHARDS", str(root))
+    results = _run_workers(_worker_sharded)
+    states = {rank: (tag, payload) for (tag, rank, payload) in results}
+    assert states[0][0] == "ok", states[0][1]
+    assert states[1][0] == "ok", states[1][1]
+    assert os.path.exists(os.path.join(states[0][1], "checkpoints",
+                                       "model_level_0.pt"))
