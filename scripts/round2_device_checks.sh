@@ -97,6 +97,44 @@ print({"multi_vs_single_max_diff": worst})
 assert worst < 1e-5, worst
 print("PASS")
 EOF
+# fused ScheduleFree step parity vs eager on a masked model:
+run schedulefree 300 python - <<'EOF2'
+import sys, torch
+sys.path.insert(0, ".")
+from turboprune_amd.ops.mask_layers import LinearMask
+from turboprune_amd.optim import ScheduleFreeSGD
+
+def one(native):
+    import turboprune_amd.ops._backend as B
+    torch.manual_seed(0)
+    m = torch.nn.Sequential(LinearMask(in_features=64, out_features=32,
+                                       bias=False)).cuda()
+    m[0].mask.bernoulli_(0.5)
+    m[0].enable_cache(torch.bfloat16)
+    opt = ScheduleFreeSGD(m.parameters(), lr=0.05, momentum=0.9,
+                          weight_decay=1e-3, model=m)
+    opt.train()
+    orig = B.use_native
+    if not native:  # force eager by hiding the dispatch
+        B.use_native = lambda *a, **k: False
+    for _ in range(5):
+        x = torch.randn(16, 64, device="cuda")
+        loss = m(x).square().mean()
+        opt.zero_grad(); loss.backward(); opt.step()
+    B.use_native = orig
+    torch.cuda.synchronize()
+    return (m[0].weight.detach().clone(),
+            m[0].weight_masked.detach().clone(),
+            m[0].mask.detach().clone())
+
+wf, cf, mask = one(True)
+we, ce, _ = one(False)
+d = (wf - we).abs().max().item()
+dc = (cf.float() - (wf * mask).to(torch.bfloat16).float()).abs().max().item()
+print({"fused_vs_eager_max_diff": d, "cache_vs_masked_w": dc})
+assert d < 1e-5 and dc == 0.0, (d, dc)
+print("PASS")
+EOF2
 # ---- promotion A/B benches (short; compare value fields) -------------
 run bench_baseline 420 python bench.py --steps 15 --warmup 5
 run bench_conv_native 420 env TURBOPRUNE_CONV=native \

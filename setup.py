@@ -36,6 +36,7 @@ sources = [
     os.path.join(CSRC, "gemm_256_8phase.hip"),
     os.path.join(CSRC, "sgd_multi.hip"),
     os.path.join(CSRC, "attention.hip"),
+    os.path.join(CSRC, "schedulefree.hip"),
 ]
 
 setup(

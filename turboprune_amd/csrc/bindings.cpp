@@ -42,6 +42,9 @@ at::Tensor attn_fwd(const at::Tensor&, const at::Tensor&,
                     const at::Tensor&, double);
 at::Tensor conv2d_implicit_gradin(const at::Tensor&, const at::Tensor&,
                                   int64_t, int64_t, int64_t, int64_t);
+void schedulefree_step_(at::Tensor, at::Tensor, const at::Tensor&,
+                        const at::Tensor&, at::Tensor, double,
+                        double, double, double);
 // fused NHWC batchnorm (batchnorm.hip)
 at::Tensor transpose2d(const at::Tensor&);
 at::Tensor conv2d_implicit_fwd(const at::Tensor&, const at::Tensor&,
@@ -94,6 +97,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "multi-tensor SGD step (cache keyed by operand data_ptrs)");
   m.def("sgd_step_multi_planned_", &turboprune::sgd_step_multi_planned_,
         "launch the multi-tensor SGD step from a prebuilt plan");
+  m.def("schedulefree_step_", &turboprune::schedulefree_step_,
+        "fused Schedule-Free SGD step (y/z update + masked-cache "
+        "rewrite in one sweep)");
   m.def("attn_fwd", &turboprune::attn_fwd,
         "EXPERIMENTAL fused flash-style attention forward, head_dim 64 "
         "(unwired; opt-in TURBOPRUNE_ATTN=native after device validation)");

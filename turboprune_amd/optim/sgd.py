@@ -154,11 +154,24 @@ class ScheduleFreeSGD(torch.optim.Optimizer):
     eval, as in the schedulefree package the reference uses."""
 
     def __init__(self, params, lr: float, momentum: float = 0.9,
-                 weight_decay: float = 0.0, warmup_steps: int = 0):
+                 weight_decay: float = 0.0, warmup_steps: int = 0,
+                 model: Optional[nn.Module] = None):
         defaults = dict(lr=lr, momentum=momentum, weight_decay=weight_decay,
                         warmup_steps=warmup_steps, k=0, train_mode=False,
                         weight_sum=0.0)
         super().__init__(params, defaults)
+        self._masked: Dict[int, Any] = {}
+        if model is not None:
+            self.attach_model(model)
+
+    def attach_model(self, model: nn.Module) -> None:
+        """weight->masked-layer map so the fused GPU step can rewrite
+        the bf16 masked-cache in the same sweep (without it the eager
+        y-iterate mutations bump every weight's _version and force a
+        mask_apply per layer per forward)."""
+        self._masked = {}
+        for _, m in masked_modules(model):
+            self._masked[id(m.weight)] = m
 
     @torch.no_grad()
     def train(self):
@@ -214,6 +227,18 @@ class ScheduleFreeSGD(torch.optim.Optimizer):
                 if "z" not in st:
                     st["z"] = p.detach().clone()
                 z = st["z"]
+                layer = self._masked.get(id(p))
+                cache = getattr(layer, "weight_masked", None) \
+                    if layer is not None else None
+                if (p.is_cuda and _backend.use_native(p)
+                        and grad.stride() == p.stride()
+                        and p.dtype == torch.float32):
+                    _backend.extension().schedulefree_step_(
+                        p, z, grad,
+                        layer.mask if layer is not None else torch.Tensor(),
+                        cache if cache is not None else torch.Tensor(),
+                        float(lr), float(beta), float(ckp1), float(wd))
+                    continue
                 if wd != 0:
                     grad = grad.add(p, alpha=wd)
                 # y -> x step then x -> y with new z
@@ -234,6 +259,7 @@ def build_optimizer(cfg: Any, model: nn.Module) -> torch.optim.Optimizer:
     if op.scheduler_type == "ScheduleFree":
         warmup = int(cfg.select("optimizer_params.warmup_steps", 0))
         return ScheduleFreeSGD(model.parameters(), lr=lr, momentum=momentum,
-                               weight_decay=wd, warmup_steps=warmup)
+                               weight_decay=wd, warmup_steps=warmup,
+                               model=model)
     return FusedMaskedSGD(model.parameters(), lr=lr, momentum=momentum,
                           weight_decay=wd, model=model)
