@@ -37,6 +37,7 @@ sources = [
     os.path.join(CSRC, "sgd_multi.hip"),
     os.path.join(CSRC, "attention.hip"),
     os.path.join(CSRC, "schedulefree.hip"),
+    os.path.join(CSRC, "conv_implicit_256.hip"),
 ]
 
 setup(

@@ -24,6 +24,13 @@
 
 namespace turboprune {
 
+// 8-phase 256x256 variant (conv_implicit_256.hip, TURBOPRUNE_CONV256=1)
+bool conv256_eligible(int Cin, int Cout, int KH, int KW);
+void conv256_launch(const __hip_bfloat16*, const __hip_bfloat16*,
+                    __hip_bfloat16*, const float*, const __hip_bfloat16*,
+                    int, int, int, int, int, int, int, int, int, int,
+                    int, int, int, bool, int, hipStream_t);
+
 using bf16x8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
@@ -248,6 +255,13 @@ at::Tensor conv2d_implicit_fwd(const at::Tensor& x_in,
   auto* yp = reinterpret_cast<__hip_bfloat16*>(y.data_ptr());
   auto* zp = reinterpret_cast<const __hip_bfloat16*>(zero_page.data_ptr());
 
+  if (conv256_eligible(Cin, Cout, KH, KW)) {
+    conv256_launch(xp, wp, yp,
+                   has_bias ? biasp.data_ptr<float>() : nullptr, zp, Nb,
+                   Hi, Wi, Cin, Cout, Ho, Wo, KH, KW, (int)stride,
+                   (int)pad, Hi, Wi, has_bias, 1, stream);
+    return y;
+  }
   if (has_bias)
     hipLaunchKernelGGL((conv3x3_fwd_kernel<__hip_bfloat16, true>), grid,
                        dim3(256), 0, stream, xp, wp, yp,
@@ -308,6 +322,12 @@ at::Tensor conv2d_implicit_gradin(const at::Tensor& gy_in,
   int new_pad = KH - 1 - (int)pad;
   // roles: "x" = gy (dilated bounds Hd,Wd; compact Ho,Wo), "Cin" = Cout,
   // "Cout" = Cin, output pixels = Hi x Wi, stride 1.
+  if (stride <= 2 && conv256_eligible(Cout, Cin, KH, KW)) {
+    conv256_launch(gp, wp, xp, nullptr, zp, Nb, Hd, Wd, Cout, Cin, Hi,
+                   Wi, KH, KW, 1, new_pad, Ho, Wo, false, (int)stride,
+                   stream);
+    return gx;
+  }
   if (stride == 1)
     hipLaunchKernelGGL((conv3x3_fwd_kernel<__hip_bfloat16, false, 1>),
                        grid, dim3(256), 0, stream, gp, wp, xp, nullptr, zp,
