@@ -72,3 +72,24 @@ def test_masked_convmask_through_fn_cpu_backend():
     F.conv2d(x, layer.mask * w2, None, 1, 1).sum().backward()
     torch.testing.assert_close(layer.weight.grad, w2.grad * layer.mask,
                                rtol=1e-4, atol=1e-4)
+
+
+def test_resnet50_dispatch_envelope():
+    """Exactly the stem (Cin=3) falls outside the native-conv shape
+    envelope on ResNet50; every other conv routes to the MFMA triple
+    once the GPU/bf16 gate opens."""
+    import torch.nn as nn
+
+    from turboprune_amd.config import compose
+    from turboprune_amd.models import build_model
+    from turboprune_amd.ops.conv_native import shape_ok
+    pm = build_model(compose("bench_resnet50_imagenet"))
+    inside, outside = [], []
+    for name, m in pm.model.named_modules():
+        if isinstance(m, nn.Conv2d):
+            ok = shape_ok(m.out_channels, m.in_channels,
+                          m.kernel_size[0], m.kernel_size[1], m.stride,
+                          m.padding, m.dilation, m.groups)
+            (inside if ok else outside).append(name)
+    assert outside == ["conv1"], outside
+    assert len(inside) == 52, len(inside)

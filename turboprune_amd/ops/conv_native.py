@@ -98,26 +98,34 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor],
                                 backend or TorchBackend)
 
 
-def native_conv_ok(x: torch.Tensor, w: torch.Tensor, stride, padding,
-                   dilation, groups: int) -> bool:
-    """True when this conv is inside the validated envelope of the MFMA
-    kernels: bf16 channels_last, square k∈{1,3} with canonical padding,
-    both channel counts %64 (Cin for fwd/wrw tiles, Cout because
-    grad_input re-enters fwd with gy's channels as the contraction)."""
-    if os.environ.get("TURBOPRUNE_CONV", "") != "native":
-        return False
-    if not (x.is_cuda and x.dtype == torch.bfloat16
-            and w.dtype == torch.bfloat16):
-        return False
+def shape_ok(cout: int, cin: int, k: int, kw: int, stride, padding,
+             dilation, groups: int) -> bool:
+    """Shape half of the dispatch envelope: square k∈{1,3} with
+    canonical padding, stride 1/2, both channel counts %64 (Cin for the
+    fwd/wrw tiles, Cout because grad_input re-enters the forward with
+    gy's channels as the contraction). On ResNet50 this admits every
+    conv except the Cin=3 stem (tests/test_conv_native_fn.py pins it)."""
     if groups != 1 or dilation != (1, 1):
         return False
-    k = w.shape[2]
-    if w.shape[3] != k or k not in (1, 3):
+    if kw != k or k not in (1, 3):
         return False
     if stride[0] != stride[1] or stride[0] not in (1, 2):
         return False
     if padding != (k // 2, k // 2):
         return False
-    if w.shape[0] % 64 or w.shape[1] % 64:
+    return cout % 64 == 0 and cin % 64 == 0
+
+
+def native_conv_ok(x: torch.Tensor, w: torch.Tensor, stride, padding,
+                   dilation, groups: int) -> bool:
+    """Full gate: opt-in env + bf16 channels_last GPU tensors +
+    shape_ok."""
+    if os.environ.get("TURBOPRUNE_CONV", "") != "native":
+        return False
+    if not (x.is_cuda and x.dtype == torch.bfloat16
+            and w.dtype == torch.bfloat16):
+        return False
+    if not shape_ok(w.shape[0], w.shape[1], w.shape[2], w.shape[3],
+                    stride, padding, dilation, groups):
         return False
     return x.is_contiguous(memory_format=torch.channels_last)
