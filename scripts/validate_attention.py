@@ -32,8 +32,12 @@ def check(ext, B, H, S, runs=1, seed=0):
     ref = ref_attn(q, k, v, scale)
     worst = 0.0
     for _ in range(runs):
-        got = ext.attn_fwd(q, k, v, scale)
+        got, lse = ext.attn_fwd(q, k, v, scale)
         worst = max(worst, (got.float() - ref).abs().max().item())
+        # lse check: logsumexp of the scaled scores
+        lref = torch.logsumexp(
+            (q.float() @ k.float().transpose(-2, -1)) * scale, dim=-1)
+        worst = max(worst, (lse - lref).abs().max().item() * 0.1)
     return worst
 
 
@@ -68,6 +72,33 @@ def main():
         print(json.dumps({"race_screen": [B, H, S],
                           "worst_err": round(worst, 5), "ok": good}))
 
+    # backward: fused kernel vs autograd through explicit attention
+    for (B, H, S) in [(1, 1, 64), (2, 3, 197)]:
+        torch.manual_seed(10 + S)
+        mk = lambda: ((torch.rand(B, H, S, 64, device="cuda") - 0.5)
+                      .bfloat16().requires_grad_())
+        q, k, v = mk(), mk(), mk()
+        scale = 1.0 / math.sqrt(64)
+        out, lse = ext.attn_fwd(q.detach(), k.detach(), v.detach(), scale)
+        s_ref = (q.float() @ k.float().transpose(-2, -1)) * scale
+        o_ref = torch.softmax(s_ref, dim=-1) @ v.float()
+        do = torch.randn_like(o_ref)
+        o_ref.backward(do)
+        dq, dk, dv = ext.attn_bwd(q.detach(), k.detach(), v.detach(), out,
+                                  do.bfloat16(), lse, scale)
+        errs = [
+            (dq.float() - q.grad.float()).abs().max().item()
+            / (q.grad.float().abs().max().item() + 1.0),
+            (dk - k.grad.float()).abs().max().item()
+            / (k.grad.float().abs().max().item() + 1.0),
+            (dv - v.grad.float()).abs().max().item()
+            / (v.grad.float().abs().max().item() + 1.0)]
+        good = max(errs) < 0.05  # relative to grad magnitude (bf16 P)
+        ok &= good
+        print(json.dumps({"bwd": [B, H, S],
+                          "max_err_dq_dk_dv": [round(e, 4) for e in errs],
+                          "ok": good}))
+
     if args.bench:
         for (B, H, S) in [(256, 6, 197), (64, 12, 197), (32, 6, 1024)]:
             mk = lambda: (torch.rand(B, H, S, 64, device="cuda") - 0.5) \
@@ -78,11 +109,20 @@ def main():
             t_sdpa = timeit(
                 lambda: torch.nn.functional
                 .scaled_dot_product_attention(q, k, v))
+            t_bwd = None
+            try:
+                out, lse = ext.attn_fwd(q, k, v, scale)
+                do = torch.randn_like(out)
+                t_bwd = round(timeit(lambda: ext.attn_bwd(
+                    q, k, v, out, do, lse, scale)), 1)
+            except Exception as e:  # noqa: BLE001
+                t_bwd = repr(e)
             flop = 4.0 * B * H * S * S * 64
             print(json.dumps({"bench": [B, H, S],
                               "ours_us": round(t_ours, 1),
                               "ours_TF": round(flop / t_ours / 1e6, 1),
-                              "sdpa_us": round(t_sdpa, 1)}))
+                              "sdpa_us": round(t_sdpa, 1),
+                              "bwd_us": t_bwd}))
 
     print("PASS" if ok else "FAIL")
     sys.exit(0 if ok else 1)

@@ -38,8 +38,11 @@ std::vector<at::Tensor> sgd_multi_plan(std::vector<at::Tensor>,
 void sgd_step_multi_planned_(const at::Tensor&, const at::Tensor&, bool,
                              bool, bool, bool, bool, double, double,
                              double);
-at::Tensor attn_fwd(const at::Tensor&, const at::Tensor&,
-                    const at::Tensor&, double);
+std::tuple<at::Tensor, at::Tensor> attn_fwd(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&, double);
+std::tuple<at::Tensor, at::Tensor, at::Tensor> attn_bwd(
+    const at::Tensor&, const at::Tensor&, const at::Tensor&,
+    const at::Tensor&, const at::Tensor&, const at::Tensor&, double);
 at::Tensor conv2d_implicit_gradin(const at::Tensor&, const at::Tensor&,
                                   int64_t, int64_t, int64_t, int64_t);
 void schedulefree_step_(at::Tensor, at::Tensor, const at::Tensor&,
@@ -101,8 +104,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused Schedule-Free SGD step (y/z update + masked-cache "
         "rewrite in one sweep)");
   m.def("attn_fwd", &turboprune::attn_fwd,
-        "EXPERIMENTAL fused flash-style attention forward, head_dim 64 "
-        "(unwired; opt-in TURBOPRUNE_ATTN=native after device validation)");
+        "EXPERIMENTAL fused flash-style attention forward -> (O, lse), "
+        "head_dim 64 (opt-in TURBOPRUNE_ATTN=native after validation)");
+  m.def("attn_bwd", &turboprune::attn_bwd,
+        "EXPERIMENTAL fused attention backward -> (dq, dk_f32, dv_f32)");
   m.def("kth_smallest", &turboprune::kth_smallest,
         "radix-select k-th smallest of a 1-D fp32 tensor");
   m.def("ce_fwd", &turboprune::ce_fwd, "fused CE forward -> (loss, lse)");

@@ -29,15 +29,22 @@ class _FlashAttn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, scale):
         from turboprune_amd.ops._backend import extension
-        out = extension().attn_fwd(q, k, v, scale)
-        ctx.save_for_backward(q, k, v)
+        out, lse = extension().attn_fwd(q, k, v, scale)
+        ctx.save_for_backward(q, k, v, out, lse)
         ctx.scale = scale
         return out
 
     @staticmethod
     def backward(ctx, do):
-        q, k, v = ctx.saved_tensors
-        gq, gk, gv = attn_backward(q, k, v, do, ctx.scale)
+        from turboprune_amd.ops._backend import extension
+        q, k, v, out, lse = ctx.saved_tensors
+        if os.environ.get("TURBOPRUNE_ATTN_BWD", "native") == "torch":
+            gq, gk, gv = attn_backward(q, k, v, do, ctx.scale)
+        else:
+            dq, dk, dv = extension().attn_bwd(q, k, v, out,
+                                              do.contiguous(), lse,
+                                              ctx.scale)
+            gq, gk, gv = dq, dk.to(k.dtype), dv.to(v.dtype)
         return gq, gk, gv, None
 
 
