@@ -59,6 +59,38 @@ for (cin, cout, k, s, hi) in [(64, 64, 3, 1, 56), (128, 128, 3, 2, 28),
 print("PASS" if ok else "FAIL")
 sys.exit(0 if ok else 1)
 EOF
+# same conv numerics with the 8-phase 256x256 variant routed in
+# (shapes chosen to hit the Cout>=192 envelope incl. one at the edge):
+run conv_native_256 300 env TURBOPRUNE_CONV256=1 python - <<'EOF'
+import json, torch, sys, os
+sys.path.insert(0, ".")
+os.environ["TURBOPRUNE_CONV"] = "native"
+from turboprune_amd.ops import conv_native
+ok = True
+for (cin, cout, k, s, hi) in [(64, 256, 3, 1, 28), (128, 256, 3, 2, 28),
+                              (256, 512, 1, 2, 28), (64, 192, 1, 1, 28)]:
+    torch.manual_seed(k + s)
+    x = (torch.rand(16, cin, hi, hi, device="cuda") - .5).bfloat16() \
+        .to(memory_format=torch.channels_last).requires_grad_()
+    w = ((torch.rand(cout, cin, k, k, device="cuda") - .5) * .1).bfloat16() \
+        .to(memory_format=torch.channels_last).requires_grad_()
+    y = conv_native.conv2d(x, w, None, s, k // 2, conv_native.NativeBackend)
+    gy = torch.randn_like(y)
+    gx, gw = torch.autograd.grad(y, [x, w], gy)
+    x2 = x.detach().clone().requires_grad_(); w2 = w.detach().clone().requires_grad_()
+    y2 = torch.nn.functional.conv2d(x2, w2, None, s, k // 2)
+    y2.backward(gy)
+    ey = (y.float() - y2.float()).abs().max().item()
+    ex = (gx.float() - x2.grad.float()).abs().max().item()
+    ew = (gw.float() - w2.grad.float()).abs().max().item()
+    sc = max(y2.float().abs().max().item(), 1.0)
+    good = ey < .05 * sc and ex < .5 and ew < .5
+    ok &= good
+    print(json.dumps({"shape": [cin, cout, k, s, hi], "ey": round(ey, 4),
+                      "ex": round(ex, 4), "ew": round(ew, 4), "ok": good}))
+print("PASS" if ok else "FAIL")
+sys.exit(0 if ok else 1)
+EOF
 # multi-tensor SGD parity vs per-tensor on a real model step:
 run multi_sgd 300 python - <<'EOF'
 import os, sys, torch
