@@ -173,6 +173,8 @@ run bench_conv_native 420 env TURBOPRUNE_CONV=native \
     python bench.py --steps 15 --warmup 5
 run bench_conv256 420 env TURBOPRUNE_CONV=native TURBOPRUNE_CONV256=1 \
     python bench.py --steps 15 --warmup 5
+run bench_conv_wrwdb 420 env TURBOPRUNE_CONV=native TURBOPRUNE_WRW_DB=1 \
+    python bench.py --steps 15 --warmup 5
 run bench_multi_sgd 420 env TURBOPRUNE_MULTI_SGD=1 \
     python bench.py --steps 15 --warmup 5
 run bench_deit_baseline 420 python bench.py --model deit_small \

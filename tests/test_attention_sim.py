@@ -45,6 +45,7 @@ def simulate_block(q, k, v, q0, scale):
     S = q.shape[0]
     sQ = pad_rows(q[q0:q0 + BQ], BQ)
     out = torch.zeros(BQ, D)
+    lse_out = torch.zeros(BQ)
 
     for wid in range(4):
         # per-lane state
@@ -118,7 +119,7 @@ def simulate_block(q, k, v, q0, scale):
                           for l in range(64)]
                     o_acc[ni] = mfma16x16x32(p_frag[kh], torch.stack(vf),
                                              o_acc[ni])
-        # epilogue
+        # epilogue (+ row logsumexp, as the kernel emits for backward)
         for ni in range(4):
             for l in range(64):
                 for j in range(4):
@@ -126,7 +127,12 @@ def simulate_block(q, k, v, q0, scale):
                     col = ni * 16 + (l & 15)
                     denom = l_i[l, j] if l_i[l, j] > 0 else 1.0
                     out[row, col] = o_acc[ni][l, j] / denom
-    return out
+        for l in range(64):
+            for j in range(4):
+                row = wid * 16 + (l >> 4) * 4 + j
+                denom = l_i[l, j] if l_i[l, j] > 0 else 1.0
+                lse_out[row] = m_i[l, j] + math.log(denom)
+    return out, lse_out
 
 
 @pytest.mark.parametrize("s", [64, 100])
@@ -137,10 +143,13 @@ def test_attention_kernel_simulation(s):
     v = torch.randn(s, D)
     scale = 1.0 / math.sqrt(D)
     ref = torch.softmax((q @ k.t()) * scale, dim=-1) @ v
+    lref = torch.logsumexp((q @ k.t()) * scale, dim=-1)
     for q0 in range(0, s, BQ):
-        got = simulate_block(q, k, v, q0, scale)
+        got, lse = simulate_block(q, k, v, q0, scale)
         n = min(BQ, s - q0)
         torch.testing.assert_close(got[:n], ref[q0:q0 + n],
+                                   rtol=1e-4, atol=1e-4)
+        torch.testing.assert_close(lse[:n], lref[q0:q0 + n],
                                    rtol=1e-4, atol=1e-4)
 
 

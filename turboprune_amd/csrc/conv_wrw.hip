@@ -177,6 +177,147 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
   }
 }
 
+// Double-buffered variant (TURBOPRUNE_WRW_DB=1, round-2 A/B): tile
+// t+1's global loads issue before tile t's MFMAs (latency hidden under
+// compute), the ds_write transpose lands in the OTHER buffer, one
+// barrier per tile. The single-buffered kernel above stages
+// synchronously — the guide's slowest staging regime. Same addresses
+// and math; only buffering/order differ. 64 KiB dynamic LDS (2 x A+B).
+__global__ __launch_bounds__(256) void conv_wrw_db_kernel(
+    const __hip_bfloat16* __restrict__ gy, const __hip_bfloat16* __restrict__ x,
+    float* __restrict__ partial, int Nb, int Hi, int Wi, int Cin, int Cout,
+    int Ho, int Wo, int KH, int KW, int stride, int pad, int Mp, int Np,
+    int grid_n) {
+  using namespace conv_wrw;
+  extern __shared__ char smem[];
+  const int kPair = 2 * BM * BK * 2;  // A+B images of one buffer
+  auto sA = [&](int buf) -> char* { return smem + buf * kPair; };
+  auto sB = [&](int buf) -> char* { return smem + buf * kPair + BM * BK * 2; };
+
+  int wg = blockIdx.x;
+  int tile_m = (wg / grid_n) * BM;
+  int tile_n = (wg % grid_n) * BN;
+  int64_t M = (int64_t)Nb * Ho * Wo;
+
+  int lane = threadIdx.x & (kWave - 1);
+  int wid = threadIdx.x / kWave;
+  int wr = wid / WARPS_N, wc = wid % WARPS_N;
+
+  int total_kt = (int)((M + BK - 1) / BK);
+  int per = (total_kt + gridDim.y - 1) / gridDim.y;
+  int kt0 = blockIdx.y * per;
+  int kt1 = min(kt0 + per, total_kt);
+  partial += (int64_t)blockIdx.y * Mp * Np;
+
+  int op = threadIdx.x & 63;
+  int g8 = threadIdx.x >> 6;  // 0..3
+
+  // registers for one staged tile (A: gy^T rows, B: im2col^T rows)
+  __hip_bfloat16 va[4][8], vb[4][8];
+
+  auto load_tile = [&](int kt) {
+    int64_t opix0 = (int64_t)kt * BK;
+    int64_t opix = opix0 + op;
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int co0 = (g8 + it * 4) * 8;
+      if (opix < M && tile_m + co0 < Cout) {
+        *reinterpret_cast<uint4*>(va[it]) = *reinterpret_cast<const uint4*>(
+            gy + opix * Cout + tile_m + co0);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) va[it][j] = __float2bfloat16(0.f);
+      }
+      int tc0 = (g8 + it * 4) * 8;
+      bool ok = false;
+      if (opix < M && tile_n + tc0 < KH * KW * Cin) {
+        int tapci = tile_n + tc0;
+        int tap = tapci / Cin;
+        int ci = tapci % Cin;
+        int dh = tap / KW, dw = tap % KW;
+        int wo = (int)(opix % Wo);
+        int64_t r2 = opix / Wo;
+        int ho = (int)(r2 % Ho);
+        int n = (int)(r2 / Ho);
+        int hi = ho * stride - pad + dh;
+        int wi = wo * stride - pad + dw;
+        if (hi >= 0 && hi < Hi && wi >= 0 && wi < Wi) {
+          *reinterpret_cast<uint4*>(vb[it]) =
+              *reinterpret_cast<const uint4*>(
+                  x + (((int64_t)n * Hi + hi) * Wi + wi) * Cin + ci);
+          ok = true;
+        }
+      }
+      if (!ok) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vb[it][j] = __float2bfloat16(0.f);
+      }
+    }
+  };
+  auto write_tile = [&](int buf) {
+#pragma unroll
+    for (int it = 0; it < 4; ++it) {
+      int r0 = (g8 + it * 4) * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        *reinterpret_cast<__hip_bfloat16*>(sA(buf) + lds_byte(r0 + j, op)) =
+            va[it][j];
+        *reinterpret_cast<__hip_bfloat16*>(sB(buf) + lds_byte(r0 + j, op)) =
+            vb[it][j];
+      }
+    }
+  };
+
+  f32x4 acc[MREP][NREP] = {};
+  if (kt0 < kt1) {
+    load_tile(kt0);
+    write_tile(0);
+  }
+  __syncthreads();
+
+  int cur = 0;
+  for (int t = kt0; t < kt1; ++t) {
+    if (t + 1 < kt1) load_tile(t + 1);  // global loads in flight
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 a_frag[MREP], b_frag[NREP];
+      int kf = ks * 32 + (lane >> 4) * 8;
+      int rowf = lane & 15;
+#pragma unroll
+      for (int mi = 0; mi < MREP; ++mi)
+        a_frag[mi] = *reinterpret_cast<const bf16x8*>(
+            sA(cur) + lds_byte(wr * WM + mi * 16 + rowf, kf));
+#pragma unroll
+      for (int ni = 0; ni < NREP; ++ni)
+        b_frag[ni] = *reinterpret_cast<const bf16x8*>(
+            sB(cur) + lds_byte(wc * WN + ni * 16 + rowf, kf));
+#pragma unroll
+      for (int mi = 0; mi < MREP; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NREP; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    if (t + 1 < kt1) write_tile(cur ^ 1);  // other buffer: no race with
+                                           // this tile's reads
+    __syncthreads();
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MREP; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < NREP; ++ni) {
+      int col = tile_n + wc * WN + ni * 16 + (lane & 15);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = tile_m + wr * WM + mi * 16 + (lane >> 4) * 4 + j;
+        partial[(int64_t)row * Np + col] = acc[mi][ni][j];
+      }
+    }
+  }
+}
+
 __global__ void wrw_reduce_kernel(const float* __restrict__ partial,
                                   __hip_bfloat16* __restrict__ out,
                                   int64_t mn, int64_t slab_stride,
@@ -221,12 +362,34 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
                            x.options().dtype(at::kFloat));
   auto stream = at::hip::getCurrentHIPStream();
   dim3 grid(tiles, splitk);
-  hipLaunchKernelGGL(conv_wrw_kernel, grid, dim3(256), 0, stream,
-                     reinterpret_cast<const __hip_bfloat16*>(gy.data_ptr()),
-                     reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
-                     partial.data_ptr<float>(), Nb, Hi, Wi, Cin, Cout, Ho,
-                     Wo, (int)KH, (int)KW, (int)stride, (int)pad, Mp, Np,
-                     grid_n);
+  static int use_db = -1;
+  if (use_db < 0) {
+    const char* e = getenv("TURBOPRUNE_WRW_DB");
+    use_db = (e && e[0] == '1') ? 1 : 0;
+  }
+  if (use_db) {
+    static bool attr_set = false;
+    if (!attr_set) {
+      hipFuncSetAttribute(reinterpret_cast<const void*>(conv_wrw_db_kernel),
+                          hipFuncAttributeMaxDynamicSharedMemorySize,
+                          2 * 2 * BM * BK * 2);
+      attr_set = true;
+    }
+    hipLaunchKernelGGL(conv_wrw_db_kernel, grid, dim3(256),
+                       2 * 2 * BM * BK * 2, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(gy.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       partial.data_ptr<float>(), Nb, Hi, Wi, Cin, Cout,
+                       Ho, Wo, (int)KH, (int)KW, (int)stride, (int)pad, Mp,
+                       Np, grid_n);
+  } else {
+    hipLaunchKernelGGL(conv_wrw_kernel, grid, dim3(256), 0, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(gy.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       partial.data_ptr<float>(), Nb, Hi, Wi, Cin, Cout,
+                       Ho, Wo, (int)KH, (int)KW, (int)stride, (int)pad, Mp,
+                       Np, grid_n);
+  }
 
   // reduce slabs -> (Cout, K) bf16, channels_last weight memory order
   auto gw = at::empty({Cout, Cin, KH, KW},
