@@ -274,3 +274,56 @@ def test_run_experiment_sharded_loader_two_ranks(tmp_path, monkeypatch):
     assert states[1][0] == "ok", states[1][1]
     assert os.path.exists(os.path.join(states[0][1], "checkpoints",
                                        "model_level_0.pt"))
+
+
+def _worker_cyclic(rank, world, port, q):
+    """Cyclic harness under DDP: per-cycle fresh optimizer + scheduler
+    must stay rank-symmetric (same step counts, same broadcasts)."""
+    os.environ["RANK"] = str(rank)
+    os.environ["LOCAL_RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    try:
+        import tempfile
+
+        from run_experiment import run
+        from turboprune_amd.config import compose
+        from turboprune_amd.harness import CyclicPruningHarness
+
+        tmp = tempfile.mkdtemp(prefix=f"tp_cyc_{rank}_")
+        cfg = compose("bench_resnet50_imagenet", [
+            "model_params.model_name=resnet18",
+            "experiment_params.epochs_per_level=2",
+            "experiment_params.distributed=true",
+            "dataset_params.total_batch_size=8",
+            "+dataset_params.steps_per_epoch=2",
+            f"experiment_params.base_dir={tmp}/experiments",
+            f"dataset_params.data_root_dir={tmp}/data",
+            "pruning_params=pai_er_erk",
+            "pruning_params.target_sparsity=0.5",
+            "cyclic_training.num_cycles=2",
+            "cyclic_training.strategy=constant",
+        ])
+        expt_dir = run(cfg, CyclicPruningHarness)
+        q.put(("ok", rank, expt_dir if rank == 0 else ""))
+    except BaseException as e:  # noqa: BLE001
+        q.put(("err", rank, repr(e)))
+        raise
+    finally:
+        for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR",
+                  "MASTER_PORT"):
+            os.environ.pop(k, None)
+
+
+@pytest.mark.timeout(600)
+def test_cyclic_harness_distributed_two_ranks():
+    results = _run_workers(_worker_cyclic)
+    states = {rank: (tag, payload) for (tag, rank, payload) in results}
+    assert states[0][0] == "ok", states[0][1]
+    assert states[1][0] == "ok", states[1][1]
+    lv = os.path.join(states[0][1], "metrics", "level_wise_metrics",
+                      "level_0_metrics.csv")
+    assert os.path.exists(lv)
+    with open(lv) as f:
+        assert "cycle" in f.readline()
