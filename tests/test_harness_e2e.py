@@ -208,7 +208,7 @@ def test_cyclic_driver_main(tmp_path, monkeypatch):
         "--config-name=cifar10_er_erk",
         f"experiment_params.base_dir={tmp_path}",
         "experiment_params.epochs_per_level=2",
-        "dataset_params.batch_size=32",
+        "dataset_params.total_batch_size=32",
         "+dataset_params.synthetic_size=64",
         "cyclic_training.num_cycles=2",
         "cyclic_training.strategy=constant",

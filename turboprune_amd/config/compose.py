@@ -209,7 +209,9 @@ def compose(config_name: str,
         cfg.merge(top)
 
     for key, val, add in kv_overrides:
-        cfg.set_path(key, val, allow_new=True if add else True)
+        # hydra semantics: plain key=value must name an existing key;
+        # '+key=value' is the explicit add syntax
+        cfg.set_path(key, val, allow_new=add)
     return cfg
 
 

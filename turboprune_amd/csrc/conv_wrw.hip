@@ -318,8 +318,13 @@ __global__ __launch_bounds__(256) void conv_wrw_db_kernel(
   }
 }
 
+// fp32 output: the split-K accumulation is fp32 and the consumer is the
+// fp32 master-weight gradient — rounding to bf16 here would be a
+// systematic numerics divergence vs the reference autocast path
+// (ADVICE r01). The output is only Cout x (KH*KW*Cin), so the extra
+// bytes are negligible next to the partial slabs.
 __global__ void wrw_reduce_kernel(const float* __restrict__ partial,
-                                  __hip_bfloat16* __restrict__ out,
+                                  float* __restrict__ out,
                                   int64_t mn, int64_t slab_stride,
                                   int slabs, int Np, int N_real,
                                   int64_t out_cols) {
@@ -331,7 +336,7 @@ __global__ void wrw_reduce_kernel(const float* __restrict__ partial,
     float v = 0.f;
     for (int s = 0; s < slabs; ++s)
       v += partial[(int64_t)s * slab_stride + row * Np + col];
-    out[i] = __float2bfloat16(v);
+    out[i] = v;
   }
 }
 
@@ -391,15 +396,15 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
                        Np, grid_n);
   }
 
-  // reduce slabs -> (Cout, K) bf16, channels_last weight memory order
+  // reduce slabs -> (Cout, K) fp32, channels_last weight memory order
   auto gw = at::empty({Cout, Cin, KH, KW},
-                      gy.options().memory_format(
+                      gy.options().dtype(at::kFloat).memory_format(
                           at::MemoryFormat::ChannelsLast));
   int64_t mn = (int64_t)Cout * K;
   int rgrid = elementwise_grid(mn, kBlock, 4);
   hipLaunchKernelGGL(wrw_reduce_kernel, dim3(rgrid), dim3(kBlock), 0,
                      stream, partial.data_ptr<float>(),
-                     reinterpret_cast<__hip_bfloat16*>(gw.data_ptr()), mn,
+                     gw.data_ptr<float>(), mn,
                      (int64_t)Mp * Np, splitk, Np, K, (int64_t)K);
   return gw;
 }

@@ -36,12 +36,20 @@ def _cache_path(root: str, dataset: str, train: bool) -> str:
 
 
 def _load_or_synthesize(root: str, dataset: str, train: bool,
-                        synthetic_size: int, seed: int):
+                        synthetic_size: int, seed: int,
+                        allow_synthetic: bool = False):
     path = _cache_path(root, dataset, train)
     if os.path.exists(path):
         with FileLock(path + ".lock"):
             blob = torch.load(path, map_location="cpu", weights_only=True)
         return blob["images"], blob["labels"], False
+    if not allow_synthetic:
+        # never silently train on noise — the
+        # reference pipeline fails fast on missing data too.
+        raise FileNotFoundError(
+            f"no {dataset} cache at {path}. Provide the tensor cache, or "
+            f"opt into synthetic data (+dataset_params.synthetic_size=N "
+            f"or TURBOPRUNE_SYNTHETIC_CIFAR=1).")
     n_classes = 100 if dataset.upper() == "CIFAR100" else 10
     g = torch.Generator().manual_seed(seed + (0 if train else 1))
     images = torch.randint(0, 256, (synthetic_size, 3, 32, 32),
@@ -57,11 +65,19 @@ class CifarLoader:
                  aug: Optional[dict] = None, altflip: bool = True,
                  device: Optional[torch.device] = None,
                  synthetic_size: int = 2048, seed: int = 0,
-                 drop_last: Optional[bool] = None):
+                 drop_last: Optional[bool] = None,
+                 allow_synthetic: Optional[bool] = None):
         self.device = device or torch.device(
             "cuda" if torch.cuda.is_available() else "cpu")
+        if allow_synthetic is None:
+            allow_synthetic = (
+                os.environ.get("TURBOPRUNE_SYNTHETIC_CIFAR") == "1")
         images_u8, labels, self.synthetic = _load_or_synthesize(
-            root, dataset, train, synthetic_size, seed)
+            root, dataset, train, synthetic_size, seed, allow_synthetic)
+        if self.synthetic and train:
+            print(f"[turboprune] WARNING: no {dataset} cache under "
+                  f"{root}; using SYNTHETIC random data "
+                  f"(explicitly allowed).", flush=True)
         self.images_u8 = images_u8.to(self.device)
         self.labels = labels.to(self.device)
         self.batch_size = batch_size
@@ -143,13 +159,20 @@ class AirbenchLoaders:
         dataset = cfg.dataset_params.dataset_name
         bs = int(cfg.dataset_params.total_batch_size)
         seed = int(cfg.select("experiment_params.seed", 0))
+        explicit = (synthetic_size is not None
+                    or cfg.select("dataset_params.synthetic_size", None)
+                    is not None)
         if synthetic_size is None:
             synthetic_size = int(cfg.select("dataset_params.synthetic_size",
                                             2048))
+        # an explicitly configured synthetic_size is the opt-in; without
+        # it CifarLoader falls through to the env-var/fail-fast check
+        allow = True if explicit else None
         self.train_loader = CifarLoader(
             root, dataset, train=True, batch_size=bs,
             aug={"flip": True, "translate": 2}, altflip=True, device=device,
-            synthetic_size=synthetic_size, seed=seed)
+            synthetic_size=synthetic_size, seed=seed, allow_synthetic=allow)
         self.test_loader = CifarLoader(
             root, dataset, train=False, batch_size=bs, device=device,
-            synthetic_size=max(synthetic_size // 4, 256), seed=seed)
+            synthetic_size=max(synthetic_size // 4, 256), seed=seed,
+            allow_synthetic=allow)

@@ -14,15 +14,27 @@ modes:
 
 - ``ShardedImageNet``: a shard reader for pre-decoded uint8 tensors
   (our beton-equivalent record format: .pt shards of
-  {images: uint8 [N,3,H,W], labels: int64}), memory-mapped host side,
-  staged to device on a side stream (double-buffered) and normalized/
-  augmented on GPU. (JPEG-on-GPU decode is gated on rocJPEG, which is
-  not present in this ROCm image.)
+  {images: uint8 [N,3,H,W], labels: int64}), memory-mapped host side.
+  A background thread gathers each batch into a persistent pinned
+  staging ring and enqueues the H2D copy on a side stream, so the
+  gather+copy of batch N+1 overlaps the GPU compute of batch N; the
+  consumer only waits on the recorded copy event. (JPEG-on-GPU decode
+  is gated on rocJPEG, which is not present in this ROCm image;
+  scripts/make_shards.py builds shards from real JPEGs via CPU decode.)
+
+Unlike the reference (which crashes inside FFCV if the .beton is
+missing), loader selection here fails fast with an explicit error when
+``dataloader_type`` names a real pipeline but no shards exist — the
+synthetic path must be opted into (``dataloader_type: synthetic``,
+``dataset_params.allow_synthetic_fallback: true`` or
+``TURBOPRUNE_ALLOW_SYNTHETIC=1``) so noise is never silently trained on.
 """
 
 from __future__ import annotations
 
 import os
+import queue
+import threading
 from typing import Any, Iterator, List, Optional, Tuple
 
 import torch
@@ -37,6 +49,8 @@ IMAGENET_VAL_SIZE = 50_000
 
 class SyntheticImageNet:
     """GPU-resident synthetic ImageNet-shaped loader (train or val)."""
+
+    synthetic = True
 
     def __init__(self, batch_size: int, device: torch.device,
                  train: bool = True, image_size: int = 224,
@@ -82,14 +96,108 @@ class SyntheticImageNet:
             yield x, self.labels[idx]
 
 
+class _Prefetcher:
+    """Background gather + H2D pipeline for ShardedImageNet.
+
+    A worker thread index_selects each batch from the mmapped shard into
+    a persistent pinned staging ring (no per-batch pin_memory()
+    allocation), enqueues the async copy on a dedicated stream, and
+    hands (device_tensor, device_labels, event) to the consumer through
+    a bounded queue — so batch N+1's host gather and H2D copy run while
+    the GPU computes on batch N. The consumer only
+    ``current_stream().wait_event(event)``s.
+
+    Ring-reuse safety: before the worker overwrites slot s it host-syncs
+    slot s's previous copy event (that copy was DEPTH batches ago).
+    """
+
+    DEPTH = 2
+
+    def __init__(self, device: torch.device, batch_shape, work):
+        self.device = device
+        self.cuda = device.type == "cuda"
+        self._stop = False
+        self._q: "queue.Queue" = queue.Queue(maxsize=self.DEPTH)
+        if self.cuda:
+            self._stream = torch.cuda.Stream(device)
+            nslots = self.DEPTH + 1
+            self._imgs = [torch.empty(batch_shape, dtype=torch.uint8,
+                                      pin_memory=True)
+                          for _ in range(nslots)]
+            self._lbls = [torch.empty(batch_shape[0], dtype=torch.int64,
+                                      pin_memory=True)
+                          for _ in range(nslots)]
+            self._events = [None] * nslots
+        self._thread = threading.Thread(target=self._run, args=(work,),
+                                        daemon=True)
+        self._thread.start()
+
+    def _put(self, item) -> bool:
+        while not self._stop:
+            try:
+                self._q.put(item, timeout=0.2)
+                return True
+            except queue.Full:
+                continue
+        return False
+
+    def _run(self, work):
+        slot = 0
+        try:
+            for images, labels, idx in work:
+                if self._stop:
+                    return
+                if self.cuda:
+                    if self._events[slot] is not None:
+                        self._events[slot].synchronize()
+                    buf, lbuf = self._imgs[slot], self._lbls[slot]
+                    torch.index_select(images, 0, idx, out=buf)
+                    torch.index_select(labels, 0, idx, out=lbuf)
+                    with torch.cuda.stream(self._stream):
+                        dev = buf.to(self.device, non_blocking=True)
+                        ldev = lbuf.to(self.device, non_blocking=True)
+                        ev = torch.cuda.Event()
+                        ev.record(self._stream)
+                    self._events[slot] = ev
+                    slot = (slot + 1) % len(self._imgs)
+                    if not self._put((dev, ldev, ev)):
+                        return
+                else:
+                    if not self._put((images[idx], labels[idx], None)):
+                        return
+            self._put(None)
+        except BaseException as e:  # surfaced to the consumer
+            self._put(e)
+
+    def __iter__(self):
+        try:
+            while True:
+                item = self._q.get()
+                if item is None:
+                    return
+                if isinstance(item, BaseException):
+                    raise item
+                dev, ldev, ev = item
+                if ev is not None:
+                    torch.cuda.current_stream().wait_event(ev)
+                yield dev, ldev
+        finally:
+            self._stop = True
+
+
 class ShardedImageNet:
-    """Pre-decoded uint8 shard reader with async H2D double buffering.
+    """Pre-decoded uint8 shard reader with threaded prefetch.
 
     Shard format: ``{root}/{split}/shard_*.pt`` each a dict with
     ``images`` uint8 [N,3,H,W] and ``labels`` int64 [N]. Sharding across
     ranks: contiguous shard round-robin by rank, per-GPU batch =
-    total_batch // world (reference sharding semantics).
+    total_batch // world (reference sharding semantics,
+    utils/dataset.py:411). When there are fewer shards than ranks, the
+    shared shard is split by index range so ranks never duplicate
+    samples (train) or double-count them (val accuracy all-reduce).
     """
+
+    synthetic = False
 
     def __init__(self, root: str, split: str, batch_size: int,
                  device: torch.device, train: bool = True,
@@ -109,11 +217,24 @@ class ShardedImageNet:
         # (mmap: metadata only), derives every rank's batch budget, and
         # stops at the global minimum — deterministic, no communication.
         sizes = [self._shard_len(p) for p in all_paths]
-        if len(all_paths) < world_size:
-            self.paths = [all_paths[rank % len(all_paths)]]
-            self.steps_per_epoch = min(sizes) // batch_size
+        S = len(all_paths)
+        if S < world_size:
+            # split the shared shard by index range across its ranks
+            def rng(r):
+                i = r % S
+                share = [q for q in range(world_size) if q % S == i]
+                sub, n = share.index(r), len(share)
+                return i, sub * sizes[i] // n, (sub + 1) * sizes[i] // n
+            i, lo, hi = rng(rank)
+            self.paths = [all_paths[i]]
+            self._ranges = [(lo, hi)]
+            self.steps_per_epoch = min(
+                (rng(q)[2] - rng(q)[1]) // batch_size
+                for q in range(world_size))
         else:
             self.paths = all_paths[rank::world_size]
+            self._ranges = [(0, sizes[all_paths.index(p)])
+                            for p in self.paths]
             budgets = [sum(s // batch_size
                            for s in sizes[r::world_size])
                        for r in range(world_size)]
@@ -126,8 +247,6 @@ class ShardedImageNet:
         self.epoch = 0
         self._mean = IMAGENET_MEAN.to(device)
         self._std = IMAGENET_STD.to(device)
-        self._copy_stream = (torch.cuda.Stream(device)
-                             if device.type == "cuda" else None)
 
     @staticmethod
     def _shard_len(path: str) -> int:
@@ -135,70 +254,86 @@ class ShardedImageNet:
                           mmap=True)
         return int(blob["images"].shape[0])
 
-    def _load_shard(self, path: str):
-        blob = torch.load(path, map_location="cpu", weights_only=True,
-                          mmap=True)
-        return blob["images"], blob["labels"]
+    def _load_shard(self, si: int):
+        blob = torch.load(self.paths[si], map_location="cpu",
+                          weights_only=True, mmap=True)
+        lo, hi = self._ranges[si]
+        return blob["images"][lo:hi], blob["labels"][lo:hi]
 
-    def __iter__(self):
+    def _batches(self):
+        """(shard_images, shard_labels, batch_index) stream consumed by
+        the prefetch thread; bounded by steps_per_epoch."""
         emitted = 0
         order = list(range(len(self.paths)))
         if self.train:
             g = torch.Generator().manual_seed(self.seed + self.epoch)
             order = torch.randperm(len(order), generator=g).tolist()
         for si in order:
-            images, labels = self._load_shard(self.paths[si])
+            images, labels = self._load_shard(si)
             n = images.shape[0]
             perm = (torch.randperm(n) if self.train else torch.arange(n))
             for i in range(0, n - self.batch_size + 1, self.batch_size):
-                idx = perm[i:i + self.batch_size]
-                raw_cpu = images[idx].pin_memory() \
-                    if self.device.type == "cuda" else images[idx]
-                if self._copy_stream is not None:
-                    with torch.cuda.stream(self._copy_stream):
-                        raw = raw_cpu.to(self.device, non_blocking=True)
-                    torch.cuda.current_stream().wait_stream(self._copy_stream)
-                else:
-                    raw = raw_cpu.to(self.device)
-                flip = None
-                if self.train:
-                    flip = torch.rand(self.batch_size,
-                                      device=self.device) < 0.5
-                hs, ws = raw.shape[-2], raw.shape[-1]
-                if self.train and (hs > self.image_size
-                                   or ws > self.image_size):
-                    # FFCV-equivalent RandomResizedCrop train pipeline
-                    boxes = augment.sample_rrc_boxes(
-                        raw.shape[0], hs, ws, device=self.device)
-                    x = augment.random_resized_crop(
-                        raw, boxes, self._mean, self._std,
-                        self.image_size, flip, self.dtype)
-                elif not self.train and (hs > self.image_size
-                                         or ws > self.image_size):
-                    # center crop (val pipeline, ratio-style)
-                    t = (hs - self.image_size) // 2
-                    l = (ws - self.image_size) // 2
-                    raw = raw[..., t:t + self.image_size,
-                              l:l + self.image_size].contiguous()
-                    x = augment.normalize_u8(raw, self._mean, self._std,
-                                             self.dtype, None)
-                else:
-                    x = augment.normalize_u8(raw, self._mean, self._std,
-                                             self.dtype, flip)
-                yield x, labels[idx].to(self.device, non_blocking=True)
+                yield images, labels, perm[i:i + self.batch_size]
                 emitted += 1
                 if emitted >= self.steps_per_epoch:
-                    self.epoch += 1
                     return
+
+    def __iter__(self):
+        H = W = None
+        # peek shard 0 for the staging shape (shards share H,W)
+        blob = torch.load(self.paths[0], map_location="cpu",
+                          weights_only=True, mmap=True)
+        H, W = blob["images"].shape[-2], blob["images"].shape[-1]
+        del blob
+        pf = _Prefetcher(self.device, (self.batch_size, 3, H, W),
+                         self._batches())
+        for raw, lbl in pf:
+            flip = None
+            if self.train:
+                flip = torch.rand(self.batch_size,
+                                  device=self.device) < 0.5
+            hs, ws = raw.shape[-2], raw.shape[-1]
+            if self.train and (hs > self.image_size
+                               or ws > self.image_size):
+                # FFCV-equivalent RandomResizedCrop train pipeline
+                boxes = augment.sample_rrc_boxes(
+                    raw.shape[0], hs, ws, device=self.device)
+                x = augment.random_resized_crop(
+                    raw, boxes, self._mean, self._std,
+                    self.image_size, flip, self.dtype)
+            elif not self.train and (hs > self.image_size
+                                     or ws > self.image_size):
+                # center crop (val pipeline, ratio-style)
+                t = (hs - self.image_size) // 2
+                l = (ws - self.image_size) // 2
+                raw = raw[..., t:t + self.image_size,
+                          l:l + self.image_size].contiguous()
+                x = augment.normalize_u8(raw, self._mean, self._std,
+                                         self.dtype, None)
+            else:
+                x = augment.normalize_u8(raw, self._mean, self._std,
+                                         self.dtype, flip)
+            yield x, lbl
         self.epoch += 1
 
     def __len__(self) -> int:
         return self.steps_per_epoch
 
 
+def _fallback_allowed(cfg: Any) -> bool:
+    return bool(cfg.select("dataset_params.allow_synthetic_fallback", False)
+                or os.environ.get("TURBOPRUNE_ALLOW_SYNTHETIC") == "1")
+
+
 class ImageNetLoaders:
     """train/test pair honoring cfg.dataset_params
-    (dataloader_type: synthetic | native | ffcv-alias)."""
+    (dataloader_type: synthetic | native | ffcv-alias).
+
+    Selection is strict: a real dataloader_type with no shards on disk
+    raises unless synthetic fallback is explicitly opted into (ADVICE
+    r01: never silently train on noise). ``self.synthetic`` records
+    which path was taken; the harness surfaces it in logs.
+    """
 
     def __init__(self, cfg: Any, device: torch.device,
                  world_size: int = 1, rank: int = 0,
@@ -210,22 +345,34 @@ class ImageNetLoaders:
         kind = cfg.dataset_params.dataloader_type
         root = cfg.dataset_params.data_root_dir
         has_shards = os.path.isdir(os.path.join(str(root), "train"))
-        if kind in ("native", "ffcv", "webdataset") and has_shards:
+        if kind in ("native", "ffcv", "webdataset", "torch") and has_shards:
+            self.synthetic = False
             self.train_loader = ShardedImageNet(
                 root, "train", per_gpu, device, True, world_size, rank,
                 seed, dtype)
             self.test_loader = ShardedImageNet(
                 root, "val", per_gpu, device, False, world_size, rank,
                 seed, dtype)
-        else:
-            # synthetic fallback (no dataset on disk / dataloader_type
-            # 'synthetic'): same shapes, same GPU pipeline
-            spe = steps_per_epoch
-            self.train_loader = SyntheticImageNet(
-                per_gpu, device, True, seed=seed + rank,
-                steps_per_epoch=spe,
-                epoch_images=IMAGENET_TRAIN_SIZE // max(world_size, 1))
-            self.test_loader = SyntheticImageNet(
-                per_gpu, device, False, seed=seed + rank,
-                steps_per_epoch=min(spe or 50, 50),
-                epoch_images=IMAGENET_VAL_SIZE // max(world_size, 1))
+            return
+        if kind != "synthetic" and not _fallback_allowed(cfg):
+            raise FileNotFoundError(
+                f"dataloader_type={kind!r} but no shards under "
+                f"{os.path.join(str(root), 'train')}. Build them with "
+                f"scripts/make_shards.py, or opt into synthetic data "
+                f"(dataset_params.dataloader_type=synthetic, "
+                f"+dataset_params.allow_synthetic_fallback=true, or "
+                f"TURBOPRUNE_ALLOW_SYNTHETIC=1).")
+        self.synthetic = True
+        if kind != "synthetic" and rank == 0:
+            print(f"[turboprune] WARNING: no shards under {root}; "
+                  f"training on SYNTHETIC random data (explicitly "
+                  f"allowed by config/env).", flush=True)
+        spe = steps_per_epoch
+        self.train_loader = SyntheticImageNet(
+            per_gpu, device, True, seed=seed + rank,
+            steps_per_epoch=spe,
+            epoch_images=IMAGENET_TRAIN_SIZE // max(world_size, 1))
+        self.test_loader = SyntheticImageNet(
+            per_gpu, device, False, seed=seed + rank,
+            steps_per_epoch=min(spe or 50, 50),
+            epoch_images=IMAGENET_VAL_SIZE // max(world_size, 1))

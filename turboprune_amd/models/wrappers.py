@@ -152,6 +152,17 @@ def build_model(cfg: Any, num_classes: Optional[int] = None) -> PruneModel:
     dataset = cfg.dataset_params.dataset_name
     if num_classes is None:
         num_classes = num_classes_of(dataset)
+    # short aliases for the timm-registered local_deit_* names
+    # (reference surface: utils/deit.py:69-253)
+    if name not in _FACTORIES:
+        for prefix in ("local_", ""):
+            cand = f"{prefix}{name}_patch16_224"
+            if cand in _FACTORIES:
+                name = cand
+                break
+        else:
+            if f"local_{name}" in _FACTORIES:
+                name = f"local_{name}"
     if name not in _FACTORIES:
         raise ValueError(f"unknown model '{name}'; available: "
                          f"{available_models()}")

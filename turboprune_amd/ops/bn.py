@@ -51,10 +51,11 @@ class _FusedBN(torch.autograd.Function):
                 None, None, None, None, None, None)
 
 
-def _composed_bn_act(bn: nn.BatchNorm2d, x, residual, relu):
+def _composed_bn_act(bn: nn.BatchNorm2d, x, residual, relu,
+                     momentum: float):
     """Reference implementation (the oracle / CPU path)."""
     y = F.batch_norm(x, bn.running_mean, bn.running_var, bn.weight,
-                     bn.bias, bn.training, bn.momentum, bn.eps)
+                     bn.bias, bn.training, momentum, bn.eps)
     if residual is not None:
         y = y + residual
     if relu:
@@ -82,10 +83,14 @@ def bn_act(bn: nn.BatchNorm2d, x: torch.Tensor,
     if bn.training and bn.track_running_stats \
             and bn.num_batches_tracked is not None:
         bn.num_batches_tracked.add_(1)
-    if not use_fused:
-        return _composed_bn_act(bn, x, residual, relu)
+    # effective momentum handles momentum=None on BOTH paths (F.batch_norm
+    # raises on None)
     momentum = bn.momentum if bn.momentum is not None \
-        else 1.0 / float(bn.num_batches_tracked)
+        else (1.0 / float(bn.num_batches_tracked)
+              if bn.num_batches_tracked is not None
+              and int(bn.num_batches_tracked) > 0 else 0.1)
+    if not use_fused:
+        return _composed_bn_act(bn, x, residual, relu, momentum)
     if bn.training and torch.is_grad_enabled():
         return _FusedBN.apply(x, residual, bn.weight, bn.bias,
                               bn.running_mean, bn.running_var, True,

@@ -98,6 +98,48 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor],
                                 backend or TorchBackend)
 
 
+class MaskedConvNativeFn(torch.autograd.Function):
+    """Masked Conv2d as ONE autograd node from the fp32 master weight
+    (reference semantics: utils/mask_layers.py:25-34 under autocast).
+
+    Forward computes on the bf16 cached masked weight (zero per-forward
+    mask multiplies); backward delivers ``grad_weight = mask ⊙ wrw`` in
+    fp32 straight to the master weight — no bf16 rounding at the
+    Function boundary (the engine would silently downcast any fp32 grad
+    returned for a bf16 input, so the conv and the mask-apply must live
+    in the same node)."""
+
+    @staticmethod
+    def forward(ctx, x, weight, mask, bias, stride, pad, w_c, backend):
+        ctx.save_for_backward(x, mask, w_c)
+        ctx.stride, ctx.pad, ctx.backend = stride, pad, backend
+        ctx.bias_dtype = None if bias is None else bias.dtype
+        ctx.weight_dtype = weight.dtype
+        return backend.fwd(x, w_c, bias, stride, pad)
+
+    @staticmethod
+    def backward(ctx, gy):
+        from turboprune_amd.ops import functional as TF
+        x, mask, w_c = ctx.saved_tensors
+        stride, pad, backend = ctx.stride, ctx.pad, ctx.backend
+        gx = gw = gb = None
+        if ctx.needs_input_grad[0]:
+            gx = backend.gradin(gy, w_c, (x.shape[2], x.shape[3]),
+                                stride, pad)
+        if ctx.needs_input_grad[1]:
+            gw = backend.wrw(gy, x, w_c.shape, stride, pad)
+            gw = TF.grad_mask_apply(gw, mask, ctx.weight_dtype)
+        if ctx.bias_dtype is not None and ctx.needs_input_grad[3]:
+            gb = gy.float().sum(dim=(0, 2, 3)).to(ctx.bias_dtype)
+        return gx, gw, None, gb, None, None, None, None
+
+
+def masked_conv2d_native(x, weight, mask, bias, stride: int, pad: int,
+                         w_cache: torch.Tensor) -> torch.Tensor:
+    return MaskedConvNativeFn.apply(x, weight, mask, bias, stride, pad,
+                                    w_cache, NativeBackend)
+
+
 def shape_ok(cout: int, cin: int, k: int, kw: int, stride, padding,
              dilation, groups: int) -> bool:
     """Shape half of the dispatch envelope: square k∈{1,3} with

@@ -74,9 +74,11 @@ def masked_conv2d(x, weight, mask, bias=None, stride=1, padding=0,
 
 def masked_linear(x, weight, mask, bias=None, cache=None, compute_dtype=None):
     w = masked_weight(weight, mask, cache, compute_dtype)
-    if (w.is_cuda and w.dtype == torch.bfloat16 and x.dtype != w.dtype
-            and torch.is_autocast_enabled()):
-        x = x.to(torch.bfloat16)  # what autocast would do inside F.linear
+    if w.dtype == torch.bfloat16 and x.dtype != w.dtype:
+        # the bf16 masked-weight cache defines the compute dtype; cast
+        # the activations like autocast would inside F.linear (also
+        # required OUTSIDE autocast — F.linear raises on mixed dtypes)
+        x = x.to(torch.bfloat16)
     # TURBOPRUNE_GEMM=library routes masked linears through hipBLASLt
     # instead of the in-house MFMA GEMM (A/B measurement knob)
     if (os.environ.get("TURBOPRUNE_GEMM", "native") == "native"

@@ -136,14 +136,24 @@ class ConvMask(_MaskedMixin, nn.Conv2d):
                 if cache is not None else None,
                 self.compute_dtype)
             return y2.view(n, h, w_, self.out_channels).permute(0, 3, 1, 2)
-        w = TF.masked_weight(self.weight, self.mask, self._fresh_cache(),
-                             self.compute_dtype)
         from turboprune_amd.ops import conv_native
-        if conv_native.native_conv_ok(x, w, self.stride, self.padding,
-                                      self.dilation, self.groups):
-            return conv_native.conv2d(
-                x, w, _bias_like(self.bias, w), self.stride[0],
-                self.padding[0], conv_native.NativeBackend)
+        cache = self._fresh_cache()
+        if (cache is not None and cache.dtype == torch.bfloat16
+                and x.is_cuda):
+            # mirror autocast input-cast semantics for the custom Function
+            if x.dtype != torch.bfloat16 and torch.is_autocast_enabled():
+                x = x.to(torch.bfloat16)
+            if conv_native.native_conv_ok(x, cache, self.stride,
+                                          self.padding, self.dilation,
+                                          self.groups):
+                # one-node masked conv: fp32 master -> bf16 compute ->
+                # fp32 mask⊙wrw grad (no bf16 grad rounding)
+                return conv_native.masked_conv2d_native(
+                    x, self.weight, self.mask,
+                    _bias_like(self.bias, cache), self.stride[0],
+                    self.padding[0], cache)
+        w = TF.masked_weight(self.weight, self.mask, cache,
+                             self.compute_dtype)
         return torch.nn.functional.conv2d(
             x, w, _bias_like(self.bias, w), self.stride, self.padding,
             self.dilation, self.groups)
