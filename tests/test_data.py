@@ -218,3 +218,95 @@ def test_sharded_equal_steps_across_ranks(tmp_path):
                              device=torch.device("cpu"), train=True,
                              world_size=4, rank=rank, image_size=8)
         assert sum(1 for _ in ld) == 1  # min shard 6 // 4 = 1
+
+
+def test_make_shards_from_real_jpegs(tmp_path):
+    """Real-JPEG ingest path (VERDICT r01 missing #2): PIL-decoded JPEG
+    tree -> uint8 shards at store-size with shorter-side resize + center
+    crop -> ShardedImageNet reads them back."""
+    import subprocess
+    import sys
+
+    from PIL import Image
+    src = tmp_path / "jpegs"
+    rng = torch.Generator().manual_seed(3)
+    for ci, cls in enumerate(["n01440764", "n01443537"]):
+        d = src / cls
+        d.mkdir(parents=True)
+        for i in range(6):
+            h = int(torch.randint(40, 90, (1,), generator=rng))
+            w = int(torch.randint(40, 90, (1,), generator=rng))
+            arr = torch.randint(0, 256, (h, w, 3), dtype=torch.uint8,
+                                generator=rng).numpy()
+            Image.fromarray(arr).save(d / f"img_{i}.jpg", quality=90)
+    out = tmp_path / "shards"
+    subprocess.run([sys.executable, "scripts/make_shards.py",
+                    "--out", str(out), "--split", "train",
+                    "--src", str(src), "--store-size", "32",
+                    "--shard-size", "8", "--workers", "2"], check=True,
+                   cwd=os.path.dirname(os.path.dirname(
+                       os.path.abspath(__file__))))
+    from turboprune_amd.data import ShardedImageNet
+    loader = ShardedImageNet(str(out), "train", batch_size=4,
+                             device=torch.device("cpu"), train=True)
+    x, y = next(iter(loader))
+    assert x.shape == (4, 3, 32, 32)
+    assert y.min() >= 0 and y.max() <= 1
+
+
+def test_sharded_split_when_fewer_shards_than_ranks(tmp_path):
+    """with 1 shard and 2 ranks, the shard is split by index
+    range — ranks must not read identical data."""
+    import subprocess
+    import sys
+    subprocess.run([sys.executable, "scripts/make_shards.py",
+                    "--out", str(tmp_path), "--split", "train",
+                    "--synthetic", "64", "--image-size", "16",
+                    "--shard-size", "64"], check=True,
+                   cwd=os.path.dirname(os.path.dirname(
+                       os.path.abspath(__file__))))
+    from turboprune_amd.data import ShardedImageNet
+    l0 = ShardedImageNet(str(tmp_path), "train", 8, torch.device("cpu"),
+                         train=False, world_size=2, rank=0)
+    l1 = ShardedImageNet(str(tmp_path), "train", 8, torch.device("cpu"),
+                         train=False, world_size=2, rank=1)
+    assert l0.steps_per_epoch == l1.steps_per_epoch == 4  # 32 imgs each
+    y0 = torch.cat([y for _, y in l0])
+    y1 = torch.cat([y for _, y in l1])
+    # index ranges are disjoint halves of the same shard
+    blob = torch.load(os.path.join(str(tmp_path), "train",
+                                   "shard_00000.pt"), weights_only=True)
+    assert torch.equal(y0, blob["labels"][:32])
+    assert torch.equal(y1, blob["labels"][32:])
+
+
+def test_imagenet_loaders_fail_fast_without_shards(tmp_path):
+    """a real dataloader_type with no shards must
+    raise, not silently train on noise."""
+    import pytest
+    cfg = compose("bench_resnet50_imagenet", [
+        "dataset_params.dataloader_type=native",
+        f"dataset_params.data_root_dir={tmp_path}/nope",
+    ])
+    from turboprune_amd.data import ImageNetLoaders
+    with pytest.raises(FileNotFoundError):
+        ImageNetLoaders(cfg, torch.device("cpu"))
+    # explicit opt-in downgrades to synthetic with a loud warning
+    cfg2 = compose("bench_resnet50_imagenet", [
+        "dataset_params.dataloader_type=native",
+        f"dataset_params.data_root_dir={tmp_path}/nope",
+        "+dataset_params.allow_synthetic_fallback=true",
+        "+dataset_params.steps_per_epoch=1",
+    ])
+    pair = ImageNetLoaders(cfg2, torch.device("cpu"), steps_per_epoch=1)
+    assert pair.synthetic
+
+
+def test_cifar_loader_fail_fast(tmp_path, monkeypatch):
+    """CifarLoader without cache and without explicit synthetic opt-in
+    must raise."""
+    import pytest
+    monkeypatch.delenv("TURBOPRUNE_SYNTHETIC_CIFAR", raising=False)
+    with pytest.raises(FileNotFoundError):
+        CifarLoader(str(tmp_path / "none"), "CIFAR10", train=True,
+                    batch_size=8)

@@ -177,3 +177,33 @@ def test_patch_embed_gemm_matches_conv():
     y.sum().backward()
     assert pe.proj.weight.grad is not None
     assert torch.all(pe.proj.weight.grad[pe.proj.mask == 0] == 0)
+
+
+def test_deit_lrr_level_loop_on_gpu(tmp_path):
+    """BASELINE config 5 surface: DeiT-Small (distilled) ImageNet
+    IMP+LRR — one full level loop through the driver on synthetic
+    ImageNet-shaped data (VERDICT r01 weak #8)."""
+    from run_experiment import run
+    from turboprune_amd.config import compose
+    cfg = compose("imagenet_deit_lrr", [
+        "model_params.model_name=local_deit_small_distilled_patch16_224",
+        "experiment_params.epochs_per_level=1",
+        "experiment_params.distributed=false",
+        "dataset_params.dataloader_type=synthetic",
+        "dataset_params.total_batch_size=32",
+        "+dataset_params.steps_per_epoch=3",
+        f"experiment_params.base_dir={tmp_path}/experiments",
+        f"dataset_params.data_root_dir={tmp_path}/data",
+        "pruning_params.target_sparsity=0.5",
+        "pruning_params.prune_rate=0.3",
+    ])
+    expt_dir = run(cfg)
+    import glob
+    levels = sorted(glob.glob(os.path.join(expt_dir, "checkpoints",
+                                           "model_level_*.pt")))
+    assert len(levels) >= 2  # at least two LRR levels ran
+    sd = torch.load(levels[-1], map_location="cpu", weights_only=True)
+    masks = {k: v for k, v in sd.items() if k.endswith("mask")}
+    assert masks, "DeiT checkpoint carries mask buffers"
+    # distilled head keys present (reference: utils/deit.py:21-66)
+    assert any("head_dist" in k for k in sd)

@@ -242,10 +242,12 @@ def test_linear_fwd_bwd_matches_torch(ext):
     assert (gw.float() - ref_gw).abs().max().item() < 1.0
 
 
-def test_masked_linear_uses_gemm(ext):
+def test_masked_linear_uses_gemm(ext, monkeypatch):
     """The LinearMask forward on GPU in bf16 must route to the MFMA GEMM
-    and match the torch oracle."""
+    and match the torch oracle (TURBOPRUNE_GEMM=native forces the
+    in-house kernel regardless of the per-shape auto routing)."""
     from turboprune_amd.ops.mask_layers import LinearMask
+    monkeypatch.setenv("TURBOPRUNE_GEMM", "native")
     torch.manual_seed(8)
     layer = LinearMask(in_features=256, out_features=128, bias=True).to(DEV)
     layer.mask.bernoulli_(0.5)

@@ -79,15 +79,32 @@ def masked_linear(x, weight, mask, bias=None, cache=None, compute_dtype=None):
         # the activations like autocast would inside F.linear (also
         # required OUTSIDE autocast — F.linear raises on mixed dtypes)
         x = x.to(torch.bfloat16)
-    # TURBOPRUNE_GEMM=library routes masked linears through hipBLASLt
-    # instead of the in-house MFMA GEMM (A/B measurement knob)
-    if (os.environ.get("TURBOPRUNE_GEMM", "native") == "native"
-            and _backend.use_native(x, w) and x.dim() >= 2):
+    # TURBOPRUNE_GEMM: auto (default, per-shape routing) | native
+    # (always in-house MFMA GEMM) | library (always hipBLASLt)
+    mode = os.environ.get("TURBOPRUNE_GEMM", "auto")
+    if (mode != "library" and _backend.use_native(x, w) and x.dim() >= 2):
         ext = _backend.extension()
         if ext is not None and hasattr(ext, "masked_linear_available") \
-                and ext.masked_linear_available(x, w):
+                and ext.masked_linear_available(x, w) \
+                and (mode == "native" or _gemm_shape_native_wins(x, w)):
             return _MaskedLinearGemm.apply(x, w, bias)
     return F.linear(x, w, bias)
+
+
+def _gemm_shape_native_wins(x, w) -> bool:
+    """Per-shape routing from the round-1 PMC table
+    (profiles/r01_gemm_mfma_pmc.md): the in-house 128² MFMA GEMM beats
+    hipBLASLt on tall token-GEMMs (M ≥ tens of thousands, N,K ≤ ~2k —
+    every DeiT training shape, incl. its split-K grad_w at 2.1×) and
+    loses on big squares (8192³: 901 vs 1565 TF) and on small-M
+    classifier heads (ResNet50 fc 512×1000×2048: 48 vs 100 TF). Routing
+    at the Function level is enough: when (M,N,K) is in the token
+    regime, both backward GEMMs (M×K×N and the deep-K N×K×M) are in
+    regimes ours wins too."""
+    K = x.shape[-1]
+    M = x.numel() // K
+    N = w.shape[0]
+    return M >= 4096 and N <= 2048 and K <= 2048
 
 
 class _MaskedLinearGemm(torch.autograd.Function):
