@@ -318,6 +318,170 @@ __global__ __launch_bounds__(256) void conv_wrw_db_kernel(
   }
 }
 
+// ---- v2 (default): BK=128 opix tiles, b128-grouped LDS staging -------
+//
+// Why v1 lost to MIOpen 3x at bs512 (profiles/r02_conv_dispatch.md):
+//   - 32 scalar ds_write_b16 per thread per operand per 64-opix tile —
+//     the slowest staging regime (guide §5.4);
+//   - 16 div/mod opix decompositions per thread per tile;
+//   - split-K targeted only 384 workgroups (256-CU chip wants >>256);
+//   - 2 barriers per 64-opix tile.
+// v2: each thread loads an 8-opix x 8-row patch (8 x uint4), transposes
+// in registers, and writes 8 x ds_write_b128 (one full 16B LDS block
+// each — 8x fewer LDS instructions per element than v1); the opix
+// decomposition is done once per tile and carried incrementally
+// (no div/mod per element); BK=128 halves the barrier count; global
+// loads for tile t+1 issue before tile t's MFMAs (register prefetch,
+// single 64 KiB LDS buffer keeps 2 blocks/CU).
+namespace conv_wrw2 {
+constexpr int BM = 128, BN = 128, BK = 128;
+constexpr int WARPS_N = 2;
+constexpr int WM = 64, WN = 64;
+constexpr int MREP = 4, NREP = 4;
+// [row][k] bf16 image, 256 B rows, XOR-swizzled 16 B blocks
+TP_DEVICE int lds_byte(int row, int k) {
+  int blk = (k >> 3) ^ (row & 7);
+  return row * (BK * 2) + blk * 16 + (k & 7) * 2;
+}
+}  // namespace conv_wrw2
+
+__global__ __launch_bounds__(256) void conv_wrw2_kernel(
+    const __hip_bfloat16* __restrict__ gy,  // (M, Cout) row-major
+    const __hip_bfloat16* __restrict__ x,   // NHWC
+    float* __restrict__ partial,            // (slabs, Mp, Np)
+    int Nb, int Hi, int Wi, int Cin, int Cout, int Ho, int Wo, int KH,
+    int KW, int stride, int pad, int Mp, int Np, int grid_n) {
+  using namespace conv_wrw2;
+  __shared__ char smem[2 * BM * BK * 2];  // A + B images, 64 KiB
+  char* sA = smem;
+  char* sB = smem + BM * BK * 2;
+
+  int wg = blockIdx.x;
+  int tile_m = (wg / grid_n) * BM;  // co tile
+  int tile_n = (wg % grid_n) * BN;  // tapci tile
+  int64_t M = (int64_t)Nb * Ho * Wo;
+
+  int lane = threadIdx.x & (kWave - 1);
+  int wid = threadIdx.x / kWave;
+  int wr = wid / WARPS_N, wc = wid % WARPS_N;
+
+  int total_kt = (int)((M + BK - 1) / BK);
+  int per = (total_kt + gridDim.y - 1) / gridDim.y;
+  int kt0 = blockIdx.y * per;
+  int kt1 = min(kt0 + per, total_kt);
+  partial += (int64_t)blockIdx.y * Mp * Np;
+
+  int kgrp = threadIdx.x & 15;   // 8-opix group within the tile
+  int ro8 = threadIdx.x >> 4;    // row octet (A: co, B: tapci)
+
+  // B octet geometry is tile-invariant: tapci -> (tap, ci) -> (dh, dw)
+  int tapci = tile_n + ro8 * 8;
+  int Ktot = KH * KW * Cin;
+  bool b_row_ok = tapci < Ktot;
+  int tap = b_row_ok ? tapci / Cin : 0;
+  int ci = b_row_ok ? tapci % Cin : 0;
+  int dh = tap / KW, dw = tap % KW;
+  int co0 = tile_m + ro8 * 8;
+  bool a_row_ok = co0 < Cout;
+
+  __hip_bfloat16 va[8][8], vb[8][8];  // [j over 8 opix][octet elem]
+
+  auto load_tile = [&](int kt) {
+    int64_t base = (int64_t)kt * BK + kgrp * 8;
+    // decompose ONCE, carry incrementally over the 8 opix
+    int wo_ = (int)(base % Wo);
+    int64_t r2 = base / Wo;
+    int ho_ = (int)(r2 % Ho);
+    int n_ = (int)(r2 / Ho);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int64_t opix = base + j;
+      if (a_row_ok && opix < M) {
+        *reinterpret_cast<uint4*>(va[j]) = *reinterpret_cast<const uint4*>(
+            gy + opix * Cout + co0);
+      } else {
+        *reinterpret_cast<uint4*>(va[j]) = uint4{0, 0, 0, 0};
+      }
+      int hi = ho_ * stride - pad + dh;
+      int wi = wo_ * stride - pad + dw;
+      if (b_row_ok && opix < M && hi >= 0 && hi < Hi && wi >= 0 &&
+          wi < Wi) {
+        *reinterpret_cast<uint4*>(vb[j]) = *reinterpret_cast<const uint4*>(
+            x + (((int64_t)n_ * Hi + hi) * Wi + wi) * Cin + ci);
+      } else {
+        *reinterpret_cast<uint4*>(vb[j]) = uint4{0, 0, 0, 0};
+      }
+      if (++wo_ == Wo) {
+        wo_ = 0;
+        if (++ho_ == Ho) {
+          ho_ = 0;
+          ++n_;
+        }
+      }
+    }
+  };
+
+  auto write_tile = [&]() {
+    // transpose [8 opix][8 rows] -> per row a 16 B run of 8 opix
+    // (k = kgrp*8 is block-aligned: one ds_write_b128 per row)
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      __hip_bfloat16 ra[8], rb[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        ra[j] = va[j][c];
+        rb[j] = vb[j][c];
+      }
+      *reinterpret_cast<uint4*>(sA + lds_byte(ro8 * 8 + c, kgrp * 8)) =
+          *reinterpret_cast<const uint4*>(ra);
+      *reinterpret_cast<uint4*>(sB + lds_byte(ro8 * 8 + c, kgrp * 8)) =
+          *reinterpret_cast<const uint4*>(rb);
+    }
+  };
+
+  f32x4 acc[MREP][NREP] = {};
+  if (kt0 < kt1) load_tile(kt0);
+  for (int t = kt0; t < kt1; ++t) {
+    write_tile();
+    __syncthreads();
+    if (t + 1 < kt1) load_tile(t + 1);  // global loads fly under MFMA
+#pragma unroll
+    for (int ks = 0; ks < 4; ++ks) {
+      bf16x8 a_frag[MREP], b_frag[NREP];
+      int kf = ks * 32 + (lane >> 4) * 8;
+      int rowf = lane & 15;
+#pragma unroll
+      for (int mi = 0; mi < MREP; ++mi)
+        a_frag[mi] = *reinterpret_cast<const bf16x8*>(
+            sA + lds_byte(wr * WM + mi * 16 + rowf, kf));
+#pragma unroll
+      for (int ni = 0; ni < NREP; ++ni)
+        b_frag[ni] = *reinterpret_cast<const bf16x8*>(
+            sB + lds_byte(wc * WN + ni * 16 + rowf, kf));
+#pragma unroll
+      for (int mi = 0; mi < MREP; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NREP; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag[ni], acc[mi][ni], 0, 0, 0);
+    }
+    __syncthreads();  // LDS consumed; next write_tile may overwrite
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MREP; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < NREP; ++ni) {
+      int col = tile_n + wc * WN + ni * 16 + (lane & 15);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = tile_m + wr * WM + mi * 16 + (lane >> 4) * 4 + j;
+        partial[(int64_t)row * Np + col] = acc[mi][ni][j];
+      }
+    }
+  }
+}
+
 // fp32 output: the split-K accumulation is fp32 and the consumer is the
 // fp32 master-weight gradient — rounding to bf16 here would be a
 // systematic numerics divergence vs the reference autocast path
@@ -358,6 +522,45 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
 
   int grid_m = Mp / BM, grid_n = Np / BN;
   int tiles = grid_m * grid_n;
+
+  // kernel version: v2 default; TURBOPRUNE_WRW=1 -> v1,
+  // TURBOPRUNE_WRW_DB=1 -> v1 double-buffered (A/B knobs)
+  static int use_db = -1, use_v1 = -1;
+  if (use_db < 0) {
+    const char* e = getenv("TURBOPRUNE_WRW_DB");
+    use_db = (e && e[0] == '1') ? 1 : 0;
+    e = getenv("TURBOPRUNE_WRW");
+    use_v1 = (e && e[0] == '1') ? 1 : 0;
+  }
+  auto stream = at::hip::getCurrentHIPStream();
+
+  if (!use_db && !use_v1) {  // v2: BK=128, >=1024-workgroup split-K
+    int total_kt = (int)((M + conv_wrw2::BK - 1) / conv_wrw2::BK);
+    int splitk = 1;
+    while (tiles * splitk < 1024 && splitk * 2 <= total_kt &&
+           splitk < 256)
+      splitk *= 2;
+    auto partial = at::empty({splitk, (int64_t)Mp, (int64_t)Np},
+                             x.options().dtype(at::kFloat));
+    hipLaunchKernelGGL(conv_wrw2_kernel, dim3(tiles, splitk), dim3(256),
+                       0, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(gy.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       partial.data_ptr<float>(), Nb, Hi, Wi, Cin, Cout,
+                       Ho, Wo, (int)KH, (int)KW, (int)stride, (int)pad, Mp,
+                       Np, grid_n);
+    auto gw = at::empty({Cout, Cin, KH, KW},
+                        gy.options().dtype(at::kFloat).memory_format(
+                            at::MemoryFormat::ChannelsLast));
+    int64_t mn = (int64_t)Cout * K;
+    int rgrid = elementwise_grid(mn, kBlock, 4);
+    hipLaunchKernelGGL(wrw_reduce_kernel, dim3(rgrid), dim3(kBlock), 0,
+                       stream, partial.data_ptr<float>(),
+                       gw.data_ptr<float>(), mn,
+                       (int64_t)Mp * Np, splitk, Np, K, (int64_t)K);
+    return gw;
+  }
+
   int total_kt = (int)((M + BK - 1) / BK);
   int splitk = 1;
   while (tiles * splitk < 384 && splitk * 4 <= total_kt && splitk < 64)
@@ -365,13 +568,7 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
 
   auto partial = at::empty({splitk, (int64_t)Mp, (int64_t)Np},
                            x.options().dtype(at::kFloat));
-  auto stream = at::hip::getCurrentHIPStream();
   dim3 grid(tiles, splitk);
-  static int use_db = -1;
-  if (use_db < 0) {
-    const char* e = getenv("TURBOPRUNE_WRW_DB");
-    use_db = (e && e[0] == '1') ? 1 : 0;
-  }
   if (use_db) {
     static bool attr_set = false;
     if (!attr_set) {

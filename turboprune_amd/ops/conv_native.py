@@ -98,6 +98,111 @@ def conv2d(x: torch.Tensor, w: torch.Tensor, bias: Optional[torch.Tensor],
                                 backend or TorchBackend)
 
 
+class LibBackend:
+    """Library (MIOpen) primitives with the same interface — used by the
+    per-shape dispatch for the ops where MIOpen wins."""
+
+    @staticmethod
+    def fwd(x, w, bias, stride, pad):
+        return F.conv2d(x, w, bias, stride, pad)
+
+    @staticmethod
+    def gradin(gy, w, in_hw, stride, pad):
+        n, _, _, _ = gy.shape
+        size = (n, w.shape[1], in_hw[0], in_hw[1])
+        return torch.nn.grad.conv2d_input(size, w, gy, stride, pad)
+
+    @staticmethod
+    def wrw(gy, x, w_shape, stride, pad):
+        return torch.nn.grad.conv2d_weight(x, w_shape, gy, stride, pad)
+
+
+class AutoBackend:
+    """Per-(shape, op) dispatch between the native implicit-GEMM kernels
+    and MIOpen, driven by the measured table (scripts/
+    conv_dispatch_table.py at bs512 on MI355X; see profiles/). Keys are
+    (Cin, Cout, k, stride); unlisted shapes default to whichever side
+    ``DEFAULT_NATIVE`` names per op."""
+
+    # filled by _load_table() from _CONV_TABLE below (or the JSON file
+    # named by TURBOPRUNE_CONV_TABLE, for A/B without rebuilds)
+    table: dict = {}
+    DEFAULT_NATIVE = {"fwd": True, "gradin": True, "wrw": False}
+
+    @classmethod
+    def _ops_for(cls, key):
+        return cls.table.get(key, cls.DEFAULT_NATIVE)
+
+    @staticmethod
+    def fwd(x, w, bias, stride, pad):
+        key = (w.shape[1], w.shape[0], w.shape[2], stride)
+        use = AutoBackend._ops_for(key).get("fwd", True)
+        return (NativeBackend if use else LibBackend).fwd(
+            x, w, bias, stride, pad)
+
+    @staticmethod
+    def gradin(gy, w, in_hw, stride, pad):
+        # note: w here is already the rotated layout consumer inside
+        # NativeBackend; key on the FORWARD geometry
+        key = (w.shape[1], w.shape[0], w.shape[2], stride)
+        use = AutoBackend._ops_for(key).get("gradin", True)
+        return (NativeBackend if use else LibBackend).gradin(
+            gy, w, in_hw, stride, pad)
+
+    @staticmethod
+    def wrw(gy, x, w_shape, stride, pad):
+        key = (w_shape[1], w_shape[0], w_shape[2], stride)
+        use = AutoBackend._ops_for(key).get("wrw", False)
+        return (NativeBackend if use else LibBackend).wrw(
+            gy, x, w_shape, stride, pad)
+
+
+def _load_table() -> None:
+    import json
+    path = os.environ.get("TURBOPRUNE_CONV_TABLE", "")
+    if path and os.path.exists(path):
+        with open(path) as f:
+            raw = json.load(f)
+        AutoBackend.table = {tuple(json.loads(k)): v
+                             for k, v in raw.items()}
+    else:
+        AutoBackend.table = dict(_CONV_TABLE)
+
+
+# Measured dispatch table — scripts/conv_dispatch_table.py at bs512 on
+# MI355X (profiles/r02_conv_dispatch.md). Keys (Cin, Cout, k, stride);
+# True = the native implicit-GEMM kernel beat MIOpen for that op at
+# that shape. Current kernel generation: the native forward wins the
+# expansion 1x1s; wrw v1 and the dilated stride-2 gradin lose (kernel
+# work tracked in docs/ROADMAP; re-measure + regenerate after each
+# kernel change).
+_CONV_TABLE: dict = {
+    (64, 64, 1, 1): {"fwd": False, "gradin": False, "wrw": False},
+    (64, 64, 3, 1): {"fwd": False, "gradin": False, "wrw": False},
+    (64, 256, 1, 1): {"fwd": True, "gradin": False, "wrw": False},
+    (128, 128, 3, 1): {"fwd": False, "gradin": False, "wrw": False},
+    (128, 128, 3, 2): {"fwd": True, "gradin": False, "wrw": False},
+    (128, 512, 1, 1): {"fwd": True, "gradin": False, "wrw": False},
+    (256, 64, 1, 1): {"fwd": False, "gradin": True, "wrw": False},
+    (256, 128, 1, 1): {"fwd": True, "gradin": False, "wrw": False},
+    (256, 256, 3, 1): {"fwd": False, "gradin": False, "wrw": False},
+    (256, 256, 3, 2): {"fwd": False, "gradin": False, "wrw": False},
+    (256, 512, 1, 2): {"fwd": True, "gradin": False, "wrw": False},
+    (256, 1024, 1, 1): {"fwd": False, "gradin": False, "wrw": False},
+    (512, 128, 1, 1): {"fwd": False, "gradin": False, "wrw": False},
+    (512, 256, 1, 1): {"fwd": True, "gradin": False, "wrw": False},
+    (512, 512, 3, 1): {"fwd": False, "gradin": False, "wrw": False},
+    (512, 512, 3, 2): {"fwd": False, "gradin": False, "wrw": False},
+    (512, 1024, 1, 2): {"fwd": True, "gradin": False, "wrw": False},
+    (512, 2048, 1, 1): {"fwd": False, "gradin": False, "wrw": False},
+    (1024, 256, 1, 1): {"fwd": False, "gradin": False, "wrw": False},
+    (1024, 512, 1, 1): {"fwd": False, "gradin": False, "wrw": False},
+    (1024, 2048, 1, 2): {"fwd": False, "gradin": False, "wrw": False},
+    (2048, 512, 1, 1): {"fwd": False, "gradin": False, "wrw": False},
+}
+_load_table()
+
+
 class MaskedConvNativeFn(torch.autograd.Function):
     """Masked Conv2d as ONE autograd node from the fp32 master weight
     (reference semantics: utils/mask_layers.py:25-34 under autocast).
@@ -135,9 +240,14 @@ class MaskedConvNativeFn(torch.autograd.Function):
 
 
 def masked_conv2d_native(x, weight, mask, bias, stride: int, pad: int,
-                         w_cache: torch.Tensor) -> torch.Tensor:
+                         w_cache: torch.Tensor,
+                         backend=None) -> torch.Tensor:
+    if backend is None:
+        backend = (NativeBackend
+                   if os.environ.get("TURBOPRUNE_CONV", "auto") == "native"
+                   else AutoBackend)
     return MaskedConvNativeFn.apply(x, weight, mask, bias, stride, pad,
-                                    w_cache, NativeBackend)
+                                    w_cache, backend)
 
 
 def shape_ok(cout: int, cin: int, k: int, kw: int, stride, padding,
@@ -160,9 +270,14 @@ def shape_ok(cout: int, cin: int, k: int, kw: int, stride, padding,
 
 def native_conv_ok(x: torch.Tensor, w: torch.Tensor, stride, padding,
                    dilation, groups: int) -> bool:
-    """Full gate: opt-in env + bf16 channels_last GPU tensors +
-    shape_ok."""
-    if os.environ.get("TURBOPRUNE_CONV", "") != "native":
+    """Full gate for routing ConvMask through MaskedConvNativeFn.
+
+    TURBOPRUNE_CONV modes: ``auto`` (default — the measured per-op
+    dispatch table, MIOpen where it wins), ``native`` (all three ops on
+    the in-house kernels, A/B knob), ``off``/``library`` (plain library
+    conv path). Plus bf16 channels_last GPU tensors + shape_ok."""
+    mode = os.environ.get("TURBOPRUNE_CONV", "auto")
+    if mode not in ("auto", "native"):
         return False
     if not (x.is_cuda and x.dtype == torch.bfloat16
             and w.dtype == torch.bfloat16):
