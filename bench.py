@@ -40,8 +40,11 @@ def parse_args():
                    choices=["bf16", "fp32"])
     p.add_argument("--no-ddp", action="store_true")
     p.add_argument("--graph", action="store_true",
-                   help="capture the train step in a hipGraph and replay "
-                        "(single-GPU; lr frozen at capture value)")
+                   help="force hipGraph capture of the train step "
+                        "(default: auto-on when per-rank batch <= 128, "
+                        "the launch-bound strong-scaling regime)")
+    p.add_argument("--no-graph", action="store_true",
+                   help="disable hipGraph capture")
     return p.parse_args()
 
 
@@ -114,8 +117,15 @@ def main():
             m.set_er_mask(1.0 - args.sparsity, seed=1234 + i)
     if bf16:
         pm.enable_caches(torch.bfloat16)
-    model = wrap_ddp(pm, cfg, device) if (distributed and not args.no_ddp) \
-        else pm
+
+    from turboprune_amd.parallel import graph_step
+    want_graph = ((args.graph or graph_step.wanted(per_gpu, distributed))
+                  and not args.no_graph)
+
+    # graph mode replaces DDP with the captured flat-grad all-reduce;
+    # eager mode keeps hook-bucketed DDP (overlap wins at large batch)
+    model = pm if want_graph or not (distributed and not args.no_ddp) \
+        else wrap_ddp(pm, cfg, device)
     opt = FusedMaskedSGD(pm.parameters(), lr=0.2, momentum=0.9,
                          weight_decay=1e-4, model=pm)
 
@@ -142,44 +152,25 @@ def main():
 
     it = iter(loader)
 
-    use_graph = args.graph and not distributed
-    if use_graph:
-        # static input buffers + captured step (HIP graph replay removes
-        # ~500 host launches per step)
-        sx, sy = next(it)
-        static_x = sx.to(memory_format=torch.channels_last).clone()
-        static_y = sy.clone()
-        # warmup on a SIDE stream (capture requirement), stable grad bufs
-        side = torch.cuda.Stream()
-        side.wait_stream(torch.cuda.current_stream())
-        with torch.cuda.stream(side):
-            for _ in range(3):
-                opt.zero_grad(set_to_none=False)
-                with torch.autocast(device_type="cuda",
-                                    dtype=torch.bfloat16, enabled=bf16):
-                    out = model(static_x)
-                    l = TF.cross_entropy(out, static_y)
-                l.backward()
-                opt.step()
-        torch.cuda.current_stream().wait_stream(side)
-        del out, l  # drop autograd graph refs before capture
-        torch.cuda.synchronize()
-        graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
-            opt.zero_grad(set_to_none=False)
-            with torch.autocast(device_type="cuda", dtype=torch.bfloat16,
-                                enabled=bf16):
-                out = model(static_x)
-                loss = TF.cross_entropy(out, static_y)
-            loss.backward()
-            opt.step()
+    if want_graph:
+        # captured step (zero -> fwd -> CE -> bwd -> flat all-reduce ->
+        # fused SGD): removes ~500 host launches per step; the capture
+        # path is the same for 1 and N GPUs (graph_step.py)
+        try:
+            sx, sy = next(it)
+            gstep = graph_step.GraphedTrainStep(
+                model, opt, TF.cross_entropy,
+                sx.to(memory_format=torch.channels_last), sy, bf16)
 
-        def one_step(batch):
-            x, y = batch
-            static_x.copy_(x.to(memory_format=torch.channels_last))
-            static_y.copy_(y)
-            graph.replay()
-            return loss
+            def one_step(batch):
+                x, y = batch
+                return gstep(x.to(memory_format=torch.channels_last), y)
+        except Exception as e:  # noqa: BLE001
+            if rank == 0:
+                print(f"# hipGraph capture failed ({e!r}); eager fallback",
+                      flush=True)
+            if distributed and not args.no_ddp:
+                model = wrap_ddp(pm, cfg, device)
 
     for _ in range(args.warmup):
         loss_out = one_step(next(it))

@@ -327,3 +327,49 @@ def test_cyclic_harness_distributed_two_ranks():
     assert os.path.exists(lv)
     with open(lv) as f:
         assert "cycle" in f.readline()
+
+
+def _worker_flatgrad_parity(rank, world, port, q):
+    """FlatGradStep (the hipGraph-capture path's gradient machinery,
+    parallel/graph_step.py) must produce the same averaged grads as
+    hook-bucketed DDP."""
+    _init(rank, world, port)
+    from turboprune_amd.parallel.graph_step import FlatGradStep
+
+    torch.manual_seed(1234)
+    x = torch.randn(2 * world, 8)
+    y = torch.randn(2 * world, 4)
+    xi = x[rank * 2:(rank + 1) * 2]
+    yi = y[rank * 2:(rank + 1) * 2]
+
+    # path A: DDP
+    from turboprune_amd.parallel.ddp import wrap_ddp
+    model_a = _tiny_model(seed=0)
+    ddp = wrap_ddp(model_a)
+    torch.nn.functional.mse_loss(ddp(xi), yi).backward()
+    ga = model_a[0].weight.grad.clone()
+
+    # path B: flat buffer + single averaged all-reduce
+    model_b = _tiny_model(seed=0)
+    flat = FlatGradStep(model_b.parameters())
+    flat.zero_()
+    torch.nn.functional.mse_loss(model_b(xi), yi).backward()
+    flat.allreduce_()
+    gb = model_b[0].weight.grad.clone()
+
+    # grads must live INSIDE the flat buffer (in-place accumulation —
+    # the property graph capture depends on)
+    ptr = model_b[0].weight.grad.data_ptr()
+    inside = (flat.flat.data_ptr() <= ptr
+              < flat.flat.data_ptr() + flat.flat.numel() * 4)
+
+    q.put(("flat", rank, (ga - gb).abs().max().item(), inside))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_flatgrad_step_matches_ddp():
+    results = _run_workers(_worker_flatgrad_parity)
+    for (_, rank, diff, inside) in results:
+        assert inside, f"rank {rank}: grad escaped the flat buffer"
+        assert diff < 1e-6, f"rank {rank}: diff {diff}"
