@@ -207,3 +207,54 @@ def test_deit_lrr_level_loop_on_gpu(tmp_path):
     assert masks, "DeiT checkpoint carries mask buffers"
     # distilled head keys present (reference: utils/deit.py:21-66)
     assert any("head_dist" in k for k in sd)
+
+
+def test_conv_auto_dispatch_composed_numerics():
+    """Masked ResNet50 fwd+bwd with the auto conv dispatch vs the pure
+    library path. The model is NOT run-to-run deterministic (MIOpen stem
+    wrw atomics, measured r2b: two identical runs diverge), so the
+    library path is run twice to establish the noise floor and the auto
+    path must stay within a small multiple of it."""
+    from turboprune_amd.config import compose
+    from turboprune_amd.models import build_model
+    from turboprune_amd.ops import functional as TF
+
+    def run(mode):
+        os.environ["TURBOPRUNE_CONV"] = mode
+        try:
+            torch.manual_seed(0)
+            cfg = compose("bench_resnet50_imagenet")
+            pm = build_model(cfg).to("cuda") \
+                .to(memory_format=torch.channels_last)
+            pm.enable_caches(torch.bfloat16)
+            torch.manual_seed(1)
+            x = torch.randn(8, 3, 224, 224, device="cuda") \
+                .to(memory_format=torch.channels_last)
+            y = torch.randint(0, 1000, (8,), device="cuda")
+            with torch.autocast("cuda", torch.bfloat16):
+                loss = TF.cross_entropy(pm(x), y)
+            loss.backward()
+            torch.cuda.synchronize()
+            return loss.item(), {n: p.grad.detach().float().clone()
+                                 for n, p in pm.named_parameters()
+                                 if p.grad is not None}
+        finally:
+            os.environ.pop("TURBOPRUNE_CONV", None)
+
+    l_a, g_a = run("off")
+    l_b, g_b = run("off")
+    l_c, g_c = run("auto")
+
+    def worst(ga, gb):
+        w = 0.0
+        for n in ga:
+            scale = ga[n].abs().max().item() + 1e-3
+            w = max(w, (ga[n] - gb[n]).abs().max().item() / scale)
+        return w
+
+    floor = worst(g_a, g_b)
+    diff = worst(g_a, g_c)
+    assert abs(l_a - l_c) < 0.1, (l_a, l_c)
+    # auto dispatch must not add error beyond the model's own
+    # nondeterminism noise (generous multiplier: bf16 tie-breaks)
+    assert diff < max(8 * floor, 0.05), (diff, floor)

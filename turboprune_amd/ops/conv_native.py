@@ -116,6 +116,15 @@ class LibBackend:
     def wrw(gy, x, w_shape, stride, pad):
         return torch.nn.grad.conv2d_weight(x, w_shape, gy, stride, pad)
 
+    @staticmethod
+    def both(gy, x, w, stride, pad, need_gx, need_gw):
+        """One fused convolution_backward computing both grads (what the
+        library autograd path would do — cheaper than two calls)."""
+        gx, gw, _ = torch.ops.aten.convolution_backward(
+            gy, x, w, None, (stride, stride), (pad, pad), (1, 1), False,
+            (0, 0), 1, (need_gx, need_gw, False))
+        return gx, gw
+
 
 class AutoBackend:
     """Per-(shape, op) dispatch between the native implicit-GEMM kernels
@@ -155,6 +164,20 @@ class AutoBackend:
         use = AutoBackend._ops_for(key).get("wrw", False)
         return (NativeBackend if use else LibBackend).wrw(
             gy, x, w_shape, stride, pad)
+
+    @staticmethod
+    def both(gy, x, w, stride, pad, need_gx, need_gw):
+        key = (w.shape[1], w.shape[0], w.shape[2], stride)
+        plan = AutoBackend._ops_for(key)
+        if plan.get("gradin", True) or plan.get("wrw", False):
+            return None  # at least one native op: per-op path
+        return LibBackend.both(gy, x, w, stride, pad, need_gx, need_gw)
+
+    @staticmethod
+    def any_native(cout, cin, k, stride) -> bool:
+        plan = AutoBackend._ops_for((cin, cout, k, stride))
+        return bool(plan.get("fwd", True) or plan.get("gradin", True)
+                    or plan.get("wrw", False))
 
 
 def _load_table() -> None:
@@ -227,12 +250,22 @@ class MaskedConvNativeFn(torch.autograd.Function):
         from turboprune_amd.ops import functional as TF
         x, mask, w_c = ctx.saved_tensors
         stride, pad, backend = ctx.stride, ctx.pad, ctx.backend
+        need_gx = ctx.needs_input_grad[0]
+        need_gw = ctx.needs_input_grad[1]
         gx = gw = gb = None
-        if ctx.needs_input_grad[0]:
-            gx = backend.gradin(gy, w_c, (x.shape[2], x.shape[3]),
-                                stride, pad)
-        if ctx.needs_input_grad[1]:
-            gw = backend.wrw(gy, x, w_c.shape, stride, pad)
+        fused = None
+        if hasattr(backend, "both") and (need_gx or need_gw):
+            fused = backend.both(gy, x, w_c, stride, pad, need_gx,
+                                 need_gw)
+        if fused is not None:
+            gx, gw = fused
+        else:
+            if need_gx:
+                gx = backend.gradin(gy, w_c, (x.shape[2], x.shape[3]),
+                                    stride, pad)
+            if need_gw:
+                gw = backend.wrw(gy, x, w_c.shape, stride, pad)
+        if gw is not None:
             gw = TF.grad_mask_apply(gw, mask, ctx.weight_dtype)
         if ctx.bias_dtype is not None and ctx.needs_input_grad[3]:
             gb = gy.float().sum(dim=(0, 2, 3)).to(ctx.bias_dtype)

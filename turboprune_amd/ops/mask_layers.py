@@ -26,6 +26,7 @@ automatically); the fused SGD step keeps the cache valid itself.
 
 from __future__ import annotations
 
+import os
 from typing import Optional
 
 import torch
@@ -145,9 +146,15 @@ class ConvMask(_MaskedMixin, nn.Conv2d):
                 x = x.to(torch.bfloat16)
             if conv_native.native_conv_ok(x, cache, self.stride,
                                           self.padding, self.dilation,
-                                          self.groups):
+                                          self.groups) and (
+                    os.environ.get("TURBOPRUNE_CONV", "auto") == "native"
+                    or conv_native.AutoBackend.any_native(
+                        self.out_channels, self.in_channels,
+                        self.kernel_size[0], self.stride[0])):
                 # one-node masked conv: fp32 master -> bf16 compute ->
-                # fp32 mask⊙wrw grad (no bf16 grad rounding)
+                # fp32 mask⊙wrw grad (no bf16 grad rounding). Shapes
+                # whose measured plan is all-library skip this Function
+                # and take the plain conv path below (zero overhead).
                 return conv_native.masked_conv2d_native(
                     x, self.weight, self.mask,
                     _bias_like(self.bias, cache), self.stride[0],
