@@ -338,9 +338,16 @@ constexpr int BM = 128, BN = 128, BK = 128;
 constexpr int WARPS_N = 2;
 constexpr int WM = 64, WN = 64;
 constexpr int MREP = 4, NREP = 4;
-// [row][k] bf16 image, 256 B rows, XOR-swizzled 16 B blocks
+// [row][k] bf16 image, 256 B rows, XOR-swizzled 16 B blocks.
+// The 256 B row stride aliases the 64 LDS banks exactly, so the
+// swizzle must spread ROWS across the 16 blocks: fold row bits 0-2 AND
+// bit 3 into the block index (16 blocks per row with BK=128 — bit 3 is
+// available). Together with the rotated write order in write_tile this
+// makes both the b128 staging writes and the b128 fragment reads
+// bank-conflict-free (lanes that differ only in row hit distinct
+// blocks).
 TP_DEVICE int lds_byte(int row, int k) {
-  int blk = (k >> 3) ^ (row & 7);
+  int blk = (k >> 3) ^ (row & 7) ^ (((row >> 3) & 1) << 3);
   return row * (BK * 2) + blk * 16 + (k & 7) * 2;
 }
 }  // namespace conv_wrw2
@@ -423,9 +430,14 @@ __global__ __launch_bounds__(256) void conv_wrw2_kernel(
 
   auto write_tile = [&]() {
     // transpose [8 opix][8 rows] -> per row a 16 B run of 8 opix
-    // (k = kgrp*8 is block-aligned: one ds_write_b128 per row)
+    // (k = kgrp*8 is block-aligned: one ds_write_b128 per row).
+    // Rotated order: at unroll step i, the four same-kgrp lanes of a
+    // wave (ro8&3 = 0..3) write rows with DISTINCT row&7 = i+2*(ro8&3),
+    // so their swizzled blocks never collide (bank-conflict-free).
+    int rot = 2 * (ro8 & 3);
 #pragma unroll
-    for (int c = 0; c < 8; ++c) {
+    for (int i = 0; i < 8; ++i) {
+      int c = (i + rot) & 7;
       __hip_bfloat16 ra[8], rb[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
