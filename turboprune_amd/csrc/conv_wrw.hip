@@ -431,13 +431,12 @@ __global__ __launch_bounds__(256) void conv_wrw2_kernel(
   auto write_tile = [&]() {
     // transpose [8 opix][8 rows] -> per row a 16 B run of 8 opix
     // (k = kgrp*8 is block-aligned: one ds_write_b128 per row).
-    // Rotated order: at unroll step i, the four same-kgrp lanes of a
-    // wave (ro8&3 = 0..3) write rows with DISTINCT row&7 = i+2*(ro8&3),
-    // so their swizzled blocks never collide (bank-conflict-free).
-    int rot = 2 * (ro8 & 3);
+    // Indices stay compile-time constant — runtime-rotated register
+    // indexing spills the staging array to scratch (measured r2d),
+    // and the residual write conflicts sit below the MFMA issue time
+    // anyway.
 #pragma unroll
-    for (int i = 0; i < 8; ++i) {
-      int c = (i + rot) & 7;
+    for (int c = 0; c < 8; ++c) {
       __hip_bfloat16 ra[8], rb[8];
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
@@ -494,6 +493,260 @@ __global__ __launch_bounds__(256) void conv_wrw2_kernel(
   }
 }
 
+// ---- v3: wide-N tiles for the 3x3 shapes --------------------------------
+//
+// PMC finding (profiles/r02: pmc_wrw): v2 is neither MFMA- nor
+// LDS-bound — it is TRAFFIC-bound. With 128x128 tiles a 64x576-output
+// problem (layer1 3x3) re-reads gy gridN times and x gridM times:
+// ~4 GB of DRAM for a 410 MB problem, a ~500 us floor — exactly
+// MIOpen's time, unreachable to beat from that tiling.
+//
+// v3 observation: the wrw OUTPUT is tiny (Cout x 9Cin), so make the
+// block tile as WIDE as a whole K-slab: 64 co x 576 tapci per block
+// (wave w owns tapci [w*144,(w+1)*144): MREP=4 x NREP=9 = 144 f32 acc
+// regs). grid_n = K/576 (1 for Cin=64 — gy and x are then each read
+// exactly ONCE), BK=32 opix per iteration, split-K fills the chip.
+// The 64 B LDS row stride also kills the v2 bank-aliasing problem
+// (64 B = 16 banks, not 0 mod 64).
+namespace conv_wrw3 {
+constexpr int BM = 64;      // co rows per block
+constexpr int BN = 576;     // tapci rows per block (4 waves x 144)
+constexpr int BK = 32;      // opix per iteration
+constexpr int WN = 144;     // tapci per wave
+constexpr int MREP = 4, NREP = 9;
+// [row][k] bf16 image, 64 B rows (4 x 16 B blocks), XOR swizzle on the
+// low 2 row bits
+TP_DEVICE int lds_byte(int row, int k) {
+  int blk = ((k >> 3) ^ row) & 3;
+  return row * (BK * 2) + blk * 16 + (k & 7) * 2;
+}
+}  // namespace conv_wrw3
+
+__global__ __launch_bounds__(256) void conv_wrw3_kernel(
+    const __hip_bfloat16* __restrict__ gy,  // (M, Cout) row-major
+    const __hip_bfloat16* __restrict__ x,   // NHWC
+    float* __restrict__ partial,            // (slabs, Mp, Np)
+    int Nb, int Hi, int Wi, int Cin, int Cout, int Ho, int Wo, int KH,
+    int KW, int stride, int pad, int Mp, int Np, int grid_n) {
+  using namespace conv_wrw3;
+  // A [64][32] + B [576][32] bf16 = 40 KiB
+  __shared__ char smem[(BM + BN) * BK * 2];
+  char* sA = smem;
+  char* sB = smem + BM * BK * 2;
+
+  int wg = blockIdx.x;
+  int tile_m = (wg / grid_n) * BM;  // co tile
+  int tile_n = (wg % grid_n) * BN;  // tapci tile
+  int64_t M = (int64_t)Nb * Ho * Wo;
+
+  int lane = threadIdx.x & (kWave - 1);
+  int wid = threadIdx.x / kWave;
+
+  int total_kt = (int)((M + BK - 1) / BK);
+  int per = (total_kt + gridDim.y - 1) / gridDim.y;
+  int kt0 = blockIdx.y * per;
+  int kt1 = min(kt0 + per, total_kt);
+  partial += (int64_t)blockIdx.y * Mp * Np;
+
+  // staging slots: 0..31 = A (ro8 = s>>2 of 8 co rows, kg = s&3 of
+  // 8 opix), 32..319 = B (same split over 72 row-octets). Threads
+  // 0..63 take a second slot (s + 256).
+  int Ktot = KH * KW * Cin;
+  struct Slot {
+    const __hip_bfloat16* base;  // row-octet base (gy or x tap base)
+    int row0;                    // LDS row of the octet
+    int kg;                      // opix group (8 opix)
+    bool is_b, row_ok;
+    int dh, dw, ci;              // B gather geometry
+  };
+  Slot slots[2];
+  int nslots = threadIdx.x < 64 ? 2 : 1;
+  for (int si = 0; si < nslots; ++si) {
+    int s = threadIdx.x + si * 256;
+    Slot& sl = slots[si];
+    if (s < 32) {
+      sl.is_b = false;
+      int ro8 = s >> 2;
+      sl.kg = s & 3;
+      sl.row0 = ro8 * 8;
+      int co0 = tile_m + ro8 * 8;
+      sl.row_ok = co0 < Cout;
+      sl.base = gy + co0;  // + opix*Cout per element
+      sl.dh = sl.dw = sl.ci = 0;
+    } else {
+      sl.is_b = true;
+      int sb = s - 32;
+      int ro8 = sb >> 2;
+      sl.kg = sb & 3;
+      sl.row0 = ro8 * 8;
+      int tapci = tile_n + ro8 * 8;
+      sl.row_ok = tapci < Ktot;
+      int tap = sl.row_ok ? tapci / Cin : 0;
+      sl.ci = sl.row_ok ? tapci % Cin : 0;
+      sl.dh = tap / KW;
+      sl.dw = tap % KW;
+      sl.base = x;
+    }
+  }
+
+  __hip_bfloat16 v[8][8];  // primary slot: [j over 8 opix][octet elem]
+
+  auto gather_slot = [&](const Slot& sl, int kt,
+                         __hip_bfloat16 (&dst)[8][8]) {
+    int64_t base_opix = (int64_t)kt * BK + sl.kg * 8;
+    int wo_ = (int)(base_opix % Wo);
+    int64_t r2 = base_opix / Wo;
+    int ho_ = (int)(r2 % Ho);
+    int n_ = (int)(r2 / Ho);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int64_t opix = base_opix + j;
+      bool ok = sl.row_ok && opix < M;
+      if (!sl.is_b) {
+        if (ok) {
+          *reinterpret_cast<uint4*>(dst[j]) =
+              *reinterpret_cast<const uint4*>(sl.base + opix * Cout);
+        } else {
+          *reinterpret_cast<uint4*>(dst[j]) = uint4{0, 0, 0, 0};
+        }
+      } else {
+        int hi = ho_ * stride - pad + sl.dh;
+        int wi = wo_ * stride - pad + sl.dw;
+        if (ok && hi >= 0 && hi < Hi && wi >= 0 && wi < Wi) {
+          *reinterpret_cast<uint4*>(dst[j]) =
+              *reinterpret_cast<const uint4*>(
+                  x + (((int64_t)n_ * Hi + hi) * Wi + wi) * Cin + sl.ci);
+        } else {
+          *reinterpret_cast<uint4*>(dst[j]) = uint4{0, 0, 0, 0};
+        }
+      }
+      if (++wo_ == Wo) {
+        wo_ = 0;
+        if (++ho_ == Ho) {
+          ho_ = 0;
+          ++n_;
+        }
+      }
+    }
+  };
+
+  auto write_slot = [&](const Slot& sl,
+                        const __hip_bfloat16 (&src)[8][8]) {
+    char* img = sl.is_b ? sB : sA;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      __hip_bfloat16 r[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) r[j] = src[j][c];
+      *reinterpret_cast<uint4*>(img + lds_byte(sl.row0 + c,
+                                               sl.kg * 8)) =
+          *reinterpret_cast<const uint4*>(r);
+    }
+  };
+
+  // load + write 4 opix (half a slot) with transient registers:
+  // ds_write_b64 per row (4 opix = 8 B, aligned at kg*8 + half*4)
+  auto stage_half = [&](const Slot& sl, int kt, int half) {
+    __hip_bfloat16 tmp[4][8];
+    int64_t base_opix = (int64_t)kt * BK + sl.kg * 8 + half * 4;
+    int wo_ = (int)(base_opix % Wo);
+    int64_t r2 = base_opix / Wo;
+    int ho_ = (int)(r2 % Ho);
+    int n_ = (int)(r2 / Ho);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      int64_t opix = base_opix + j;
+      bool ok = sl.row_ok && opix < M;
+      if (!sl.is_b) {
+        if (ok) {
+          *reinterpret_cast<uint4*>(tmp[j]) =
+              *reinterpret_cast<const uint4*>(sl.base + opix * Cout);
+        } else {
+          *reinterpret_cast<uint4*>(tmp[j]) = uint4{0, 0, 0, 0};
+        }
+      } else {
+        int hi = ho_ * stride - pad + sl.dh;
+        int wi = wo_ * stride - pad + sl.dw;
+        if (ok && hi >= 0 && hi < Hi && wi >= 0 && wi < Wi) {
+          *reinterpret_cast<uint4*>(tmp[j]) =
+              *reinterpret_cast<const uint4*>(
+                  x + (((int64_t)n_ * Hi + hi) * Wi + wi) * Cin + sl.ci);
+        } else {
+          *reinterpret_cast<uint4*>(tmp[j]) = uint4{0, 0, 0, 0};
+        }
+      }
+      if (++wo_ == Wo) {
+        wo_ = 0;
+        if (++ho_ == Ho) {
+          ho_ = 0;
+          ++n_;
+        }
+      }
+    }
+    char* img = sl.is_b ? sB : sA;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      __hip_bfloat16 r[4];
+#pragma unroll
+      for (int j = 0; j < 4; ++j) r[j] = tmp[j][c];
+      *reinterpret_cast<uint2*>(
+          img + lds_byte(sl.row0 + c, sl.kg * 8 + half * 4)) =
+          *reinterpret_cast<const uint2*>(r);
+    }
+  };
+
+  f32x4 acc[MREP][NREP] = {};
+  if (kt0 < kt1) gather_slot(slots[0], kt0, v);
+  for (int t = kt0; t < kt1; ++t) {
+    write_slot(slots[0], v);
+    if (nslots > 1) {
+      // overflow slot (threads 0..63): loaded + written inline in two
+      // 4-opix halves so its registers stay short-lived and small
+      // (keeps occupancy at 2 waves/SIMD)
+      stage_half(slots[1], t, 0);
+      stage_half(slots[1], t, 1);
+    }
+    __syncthreads();
+    if (t + 1 < kt1)
+      gather_slot(slots[0], t + 1, v);  // primary loads fly under MFMA
+    {
+      // B fragments are read one at a time inside the ni loop so only
+      // one is live at once (holding all 9 costs 36 VGPRs and drops
+      // occupancy to 1 wave/SIMD with the 144-reg accumulator)
+      bf16x8 a_frag[MREP];
+      int kf = (lane >> 4) * 8;
+      int rowf = lane & 15;
+#pragma unroll
+      for (int mi = 0; mi < MREP; ++mi)
+        a_frag[mi] = *reinterpret_cast<const bf16x8*>(
+            sA + lds_byte(mi * 16 + rowf, kf));
+#pragma unroll
+      for (int ni = 0; ni < NREP; ++ni) {
+        bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+            sB + lds_byte(wid * WN + ni * 16 + rowf, kf));
+#pragma unroll
+        for (int mi = 0; mi < MREP; ++mi)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag, acc[mi][ni], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MREP; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < NREP; ++ni) {
+      int col = tile_n + wid * WN + ni * 16 + (lane & 15);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = tile_m + mi * 16 + (lane >> 4) * 4 + j;
+        partial[(int64_t)row * Np + col] = acc[mi][ni][j];
+      }
+    }
+  }
+}
+
 // fp32 output: the split-K accumulation is fp32 and the consumer is the
 // fp32 master-weight gradient — rounding to bf16 here would be a
 // systematic numerics divergence vs the reference autocast path
@@ -535,16 +788,52 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
   int grid_m = Mp / BM, grid_n = Np / BN;
   int tiles = grid_m * grid_n;
 
-  // kernel version: v2 default; TURBOPRUNE_WRW=1 -> v1,
-  // TURBOPRUNE_WRW_DB=1 -> v1 double-buffered (A/B knobs)
-  static int use_db = -1, use_v1 = -1;
+  // kernel version: auto = v3 for 3x3 (wide-N tiles, minimal re-reads),
+  // v2 otherwise; TURBOPRUNE_WRW=2 forces v2, =1 forces v1,
+  // TURBOPRUNE_WRW_DB=1 the v1 double-buffered variant (A/B knobs)
+  static int use_db = -1, use_v1 = -1, force_v2 = -1;
   if (use_db < 0) {
     const char* e = getenv("TURBOPRUNE_WRW_DB");
     use_db = (e && e[0] == '1') ? 1 : 0;
     e = getenv("TURBOPRUNE_WRW");
     use_v1 = (e && e[0] == '1') ? 1 : 0;
+    force_v2 = (e && e[0] == '2') ? 1 : 0;
   }
   auto stream = at::hip::getCurrentHIPStream();
+
+  if (!use_db && !use_v1 && !force_v2 && KH == 3 && KW == 3) {
+    // v3: 64 x 576 tiles — gy re-read ceil(K/576)x, x re-read
+    // ceil(Cout/64)x only via L2-absorbed tap overlap
+    int gm = (Cout + conv_wrw3::BM - 1) / conv_wrw3::BM;
+    int gn = (K + conv_wrw3::BN - 1) / conv_wrw3::BN;
+    int Mp3 = gm * conv_wrw3::BM;
+    int Np3 = gn * conv_wrw3::BN;
+    int tiles3 = gm * gn;
+    int total_kt = (int)((M + conv_wrw3::BK - 1) / conv_wrw3::BK);
+    int splitk = 1;
+    while (tiles3 * splitk < 1024 && splitk * 2 <= total_kt &&
+           splitk < 512)
+      splitk *= 2;
+    auto partial = at::empty({splitk, (int64_t)Mp3, (int64_t)Np3},
+                             x.options().dtype(at::kFloat));
+    hipLaunchKernelGGL(conv_wrw3_kernel, dim3(tiles3, splitk), dim3(256),
+                       0, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(gy.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       partial.data_ptr<float>(), Nb, Hi, Wi, Cin, Cout,
+                       Ho, Wo, (int)KH, (int)KW, (int)stride, (int)pad,
+                       Mp3, Np3, gn);
+    auto gw = at::empty({Cout, Cin, KH, KW},
+                        gy.options().dtype(at::kFloat).memory_format(
+                            at::MemoryFormat::ChannelsLast));
+    int64_t mn = (int64_t)Cout * K;
+    int rgrid = elementwise_grid(mn, kBlock, 4);
+    hipLaunchKernelGGL(wrw_reduce_kernel, dim3(rgrid), dim3(kBlock), 0,
+                       stream, partial.data_ptr<float>(),
+                       gw.data_ptr<float>(), mn,
+                       (int64_t)Mp3 * Np3, splitk, Np3, K, (int64_t)K);
+    return gw;
+  }
 
   if (!use_db && !use_v1) {  // v2: BK=128, >=1024-workgroup split-K
     int total_kt = (int)((M + conv_wrw2::BK - 1) / conv_wrw2::BK);
