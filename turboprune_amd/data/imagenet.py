@@ -255,7 +255,17 @@ class ShardedImageNet:
         return int(blob["images"].shape[0])
 
     def _load_shard(self, si: int):
-        blob = torch.load(self.paths[si], map_location="cpu",
+        path = self.paths[si]
+        try:
+            # async whole-file readahead: the mmap gather otherwise
+            # page-faults row by row at disk latency (measured r2d:
+            # ~450 MB/s vs ~17 GB/s warm)
+            fd = os.open(path, os.O_RDONLY)
+            os.posix_fadvise(fd, 0, 0, os.POSIX_FADV_WILLNEED)
+            os.close(fd)
+        except OSError:
+            pass
+        blob = torch.load(path, map_location="cpu",
                           weights_only=True, mmap=True)
         lo, hi = self._ranges[si]
         return blob["images"][lo:hi], blob["labels"][lo:hi]
