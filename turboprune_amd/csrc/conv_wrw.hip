@@ -550,50 +550,41 @@ __global__ __launch_bounds__(256) void conv_wrw3_kernel(
 
   // staging slots: 0..31 = A (ro8 = s>>2 of 8 co rows, kg = s&3 of
   // 8 opix), 32..319 = B (same split over 72 row-octets). Threads
-  // 0..63 take a second slot (s + 256).
+  // 0..63 take a second slot (s + 256). All slot state lives in plain
+  // scalar locals and the staging arrays are captured (not passed) by
+  // the lambdas — aggregates passed by reference defeat SROA and land
+  // the hot-loop state in scratch (measured r2e: 120 scratch ops, 6x
+  // slowdown).
   int Ktot = KH * KW * Cin;
-  struct Slot {
-    const __hip_bfloat16* base;  // row-octet base (gy or x tap base)
-    int row0;                    // LDS row of the octet
-    int kg;                      // opix group (8 opix)
-    bool is_b, row_ok;
-    int dh, dw, ci;              // B gather geometry
-  };
-  Slot slots[2];
-  int nslots = threadIdx.x < 64 ? 2 : 1;
-  for (int si = 0; si < nslots; ++si) {
-    int s = threadIdx.x + si * 256;
-    Slot& sl = slots[si];
-    if (s < 32) {
-      sl.is_b = false;
-      int ro8 = s >> 2;
-      sl.kg = s & 3;
-      sl.row0 = ro8 * 8;
-      int co0 = tile_m + ro8 * 8;
-      sl.row_ok = co0 < Cout;
-      sl.base = gy + co0;  // + opix*Cout per element
-      sl.dh = sl.dw = sl.ci = 0;
-    } else {
-      sl.is_b = true;
-      int sb = s - 32;
-      int ro8 = sb >> 2;
-      sl.kg = sb & 3;
-      sl.row0 = ro8 * 8;
-      int tapci = tile_n + ro8 * 8;
-      sl.row_ok = tapci < Ktot;
-      int tap = sl.row_ok ? tapci / Cin : 0;
-      sl.ci = sl.row_ok ? tapci % Cin : 0;
-      sl.dh = tap / KW;
-      sl.dw = tap % KW;
-      sl.base = x;
-    }
-  }
+
+  // decode slot s -> (is_b, row0, kg, row_ok, src geometry)
+  int s0 = threadIdx.x;
+  bool b0_is_b = s0 >= 32;
+  int p0_ro8 = b0_is_b ? (s0 - 32) >> 2 : s0 >> 2;
+  int p0_kg = b0_is_b ? (s0 - 32) & 3 : s0 & 3;
+  int p0_row0 = p0_ro8 * 8;
+  int p0_tapci = tile_n + p0_row0;
+  bool p0_ok = b0_is_b ? (p0_tapci < Ktot) : (tile_m + p0_row0 < Cout);
+  int p0_tap = (b0_is_b && p0_ok) ? p0_tapci / Cin : 0;
+  int p0_ci = (b0_is_b && p0_ok) ? p0_tapci % Cin : 0;
+  int p0_dh = p0_tap / KW, p0_dw = p0_tap % KW;
+  const __hip_bfloat16* p0_gy = gy + tile_m + p0_row0;
+
+  int s1 = threadIdx.x + 256;      // overflow slot (threads 0..63): B
+  bool has1 = threadIdx.x < 64;
+  int p1_ro8 = (s1 - 32) >> 2;
+  int p1_kg = (s1 - 32) & 3;
+  int p1_row0 = p1_ro8 * 8;
+  int p1_tapci = tile_n + p1_row0;
+  bool p1_ok = has1 && p1_tapci < Ktot;
+  int p1_tap = p1_ok ? p1_tapci / Cin : 0;
+  int p1_ci = p1_ok ? p1_tapci % Cin : 0;
+  int p1_dh = p1_tap / KW, p1_dw = p1_tap % KW;
 
   __hip_bfloat16 v[8][8];  // primary slot: [j over 8 opix][octet elem]
 
-  auto gather_slot = [&](const Slot& sl, int kt,
-                         __hip_bfloat16 (&dst)[8][8]) {
-    int64_t base_opix = (int64_t)kt * BK + sl.kg * 8;
+  auto load_primary = [&](int kt) {
+    int64_t base_opix = (int64_t)kt * BK + p0_kg * 8;
     int wo_ = (int)(base_opix % Wo);
     int64_t r2 = base_opix / Wo;
     int ho_ = (int)(r2 % Ho);
@@ -601,23 +592,23 @@ __global__ __launch_bounds__(256) void conv_wrw3_kernel(
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       int64_t opix = base_opix + j;
-      bool ok = sl.row_ok && opix < M;
-      if (!sl.is_b) {
+      bool ok = p0_ok && opix < M;
+      if (!b0_is_b) {
         if (ok) {
-          *reinterpret_cast<uint4*>(dst[j]) =
-              *reinterpret_cast<const uint4*>(sl.base + opix * Cout);
+          *reinterpret_cast<uint4*>(v[j]) =
+              *reinterpret_cast<const uint4*>(p0_gy + opix * Cout);
         } else {
-          *reinterpret_cast<uint4*>(dst[j]) = uint4{0, 0, 0, 0};
+          *reinterpret_cast<uint4*>(v[j]) = uint4{0, 0, 0, 0};
         }
       } else {
-        int hi = ho_ * stride - pad + sl.dh;
-        int wi = wo_ * stride - pad + sl.dw;
+        int hi = ho_ * stride - pad + p0_dh;
+        int wi = wo_ * stride - pad + p0_dw;
         if (ok && hi >= 0 && hi < Hi && wi >= 0 && wi < Wi) {
-          *reinterpret_cast<uint4*>(dst[j]) =
+          *reinterpret_cast<uint4*>(v[j]) =
               *reinterpret_cast<const uint4*>(
-                  x + (((int64_t)n_ * Hi + hi) * Wi + wi) * Cin + sl.ci);
+                  x + (((int64_t)n_ * Hi + hi) * Wi + wi) * Cin + p0_ci);
         } else {
-          *reinterpret_cast<uint4*>(dst[j]) = uint4{0, 0, 0, 0};
+          *reinterpret_cast<uint4*>(v[j]) = uint4{0, 0, 0, 0};
         }
       }
       if (++wo_ == Wo) {
@@ -630,25 +621,22 @@ __global__ __launch_bounds__(256) void conv_wrw3_kernel(
     }
   };
 
-  auto write_slot = [&](const Slot& sl,
-                        const __hip_bfloat16 (&src)[8][8]) {
-    char* img = sl.is_b ? sB : sA;
+  auto write_primary = [&]() {
+    char* img = b0_is_b ? sB : sA;
 #pragma unroll
     for (int c = 0; c < 8; ++c) {
       __hip_bfloat16 r[8];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) r[j] = src[j][c];
-      *reinterpret_cast<uint4*>(img + lds_byte(sl.row0 + c,
-                                               sl.kg * 8)) =
+      for (int j = 0; j < 8; ++j) r[j] = v[j][c];
+      *reinterpret_cast<uint4*>(img + lds_byte(p0_row0 + c, p0_kg * 8)) =
           *reinterpret_cast<const uint4*>(r);
     }
   };
 
-  // load + write 4 opix (half a slot) with transient registers:
-  // ds_write_b64 per row (4 opix = 8 B, aligned at kg*8 + half*4)
-  auto stage_half = [&](const Slot& sl, int kt, int half) {
+  // overflow slot: load + write 4 opix at a time, transient registers
+  auto stage1_half = [&](int kt, int half) {
     __hip_bfloat16 tmp[4][8];
-    int64_t base_opix = (int64_t)kt * BK + sl.kg * 8 + half * 4;
+    int64_t base_opix = (int64_t)kt * BK + p1_kg * 8 + half * 4;
     int wo_ = (int)(base_opix % Wo);
     int64_t r2 = base_opix / Wo;
     int ho_ = (int)(r2 % Ho);
@@ -656,24 +644,15 @@ __global__ __launch_bounds__(256) void conv_wrw3_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       int64_t opix = base_opix + j;
-      bool ok = sl.row_ok && opix < M;
-      if (!sl.is_b) {
-        if (ok) {
-          *reinterpret_cast<uint4*>(tmp[j]) =
-              *reinterpret_cast<const uint4*>(sl.base + opix * Cout);
-        } else {
-          *reinterpret_cast<uint4*>(tmp[j]) = uint4{0, 0, 0, 0};
-        }
+      int hi = ho_ * stride - pad + p1_dh;
+      int wi = wo_ * stride - pad + p1_dw;
+      if (p1_ok && opix < M && hi >= 0 && hi < Hi && wi >= 0 &&
+          wi < Wi) {
+        *reinterpret_cast<uint4*>(tmp[j]) =
+            *reinterpret_cast<const uint4*>(
+                x + (((int64_t)n_ * Hi + hi) * Wi + wi) * Cin + p1_ci);
       } else {
-        int hi = ho_ * stride - pad + sl.dh;
-        int wi = wo_ * stride - pad + sl.dw;
-        if (ok && hi >= 0 && hi < Hi && wi >= 0 && wi < Wi) {
-          *reinterpret_cast<uint4*>(tmp[j]) =
-              *reinterpret_cast<const uint4*>(
-                  x + (((int64_t)n_ * Hi + hi) * Wi + wi) * Cin + sl.ci);
-        } else {
-          *reinterpret_cast<uint4*>(tmp[j]) = uint4{0, 0, 0, 0};
-        }
+        *reinterpret_cast<uint4*>(tmp[j]) = uint4{0, 0, 0, 0};
       }
       if (++wo_ == Wo) {
         wo_ = 0;
@@ -683,32 +662,29 @@ __global__ __launch_bounds__(256) void conv_wrw3_kernel(
         }
       }
     }
-    char* img = sl.is_b ? sB : sA;
 #pragma unroll
     for (int c = 0; c < 8; ++c) {
       __hip_bfloat16 r[4];
 #pragma unroll
       for (int j = 0; j < 4; ++j) r[j] = tmp[j][c];
       *reinterpret_cast<uint2*>(
-          img + lds_byte(sl.row0 + c, sl.kg * 8 + half * 4)) =
+          sB + lds_byte(p1_row0 + c, p1_kg * 8 + half * 4)) =
           *reinterpret_cast<const uint2*>(r);
     }
   };
 
   f32x4 acc[MREP][NREP] = {};
-  if (kt0 < kt1) gather_slot(slots[0], kt0, v);
+  if (kt0 < kt1) load_primary(kt0);
   for (int t = kt0; t < kt1; ++t) {
-    write_slot(slots[0], v);
-    if (nslots > 1) {
+    write_primary();
+    if (has1) {
       // overflow slot (threads 0..63): loaded + written inline in two
       // 4-opix halves so its registers stay short-lived and small
-      // (keeps occupancy at 2 waves/SIMD)
-      stage_half(slots[1], t, 0);
-      stage_half(slots[1], t, 1);
+      stage1_half(t, 0);
+      stage1_half(t, 1);
     }
     __syncthreads();
-    if (t + 1 < kt1)
-      gather_slot(slots[0], t + 1, v);  // primary loads fly under MFMA
+    if (t + 1 < kt1) load_primary(t + 1);  // loads fly under MFMA
     {
       // B fragments are read one at a time inside the ni loop so only
       // one is live at once (holding all 9 costs 36 VGPRs and drops
