@@ -583,20 +583,53 @@ __global__ __launch_bounds__(256) void conv_wrw3_kernel(
 
   __hip_bfloat16 v[8][8];  // primary slot: [j over 8 opix][octet elem]
 
-  auto load_primary = [&](int kt) {
-    int64_t base_opix = (int64_t)kt * BK + p0_kg * 8;
-    int wo_ = (int)(base_opix % Wo);
-    int64_t r2 = base_opix / Wo;
-    int ho_ = (int)(r2 % Ho);
-    int n_ = (int)(r2 / Ho);
+  // All coordinate state is 32-bit and advanced INCREMENTALLY across
+  // iterations: the original per-tile int64 divmods expanded to ~150
+  // soft-division instructions each and made the loop body ~3000 ISA
+  // instructions against 36 MFMAs (r2f measurement). One divmod at
+  // kt0, then carry chains only. M, opix and byte offsets all fit in
+  // 32 bits for every supported shape (M <= 512*112*112 < 2^31).
+  int Mi = (int)M;
+
+  struct Coord {
+    int wo, ho, n;
+  };
+  auto coord_init = [&](int opix0) -> Coord {
+    Coord c;
+    c.wo = opix0 % Wo;
+    int r2 = opix0 / Wo;
+    c.ho = r2 % Ho;
+    c.n = r2 / Ho;
+    return c;
+  };
+  auto coord_step = [&](Coord& c, int by) {
+    c.wo += by;
+    while (c.wo >= Wo) {
+      c.wo -= Wo;
+      if (++c.ho == Ho) {
+        c.ho = 0;
+        ++c.n;
+      }
+    }
+  };
+
+  int p0_base = kt0 * BK + p0_kg * 8;  // this slot's first opix
+  // clamped inits keep the divmod benign for out-of-range slots; every
+  // load is guarded by opix < Mi anyway
+  Coord c0 = coord_init(p0_base < Mi ? p0_base : 0);
+  int p1_base = kt0 * BK + p1_kg * 8;
+  Coord c1 = coord_init((has1 && p1_base < Mi) ? p1_base : 0);
+
+  auto load_primary = [&](int opix_b, const Coord& cb) {
+    int wo_ = cb.wo, ho_ = cb.ho, n_ = cb.n;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      int64_t opix = base_opix + j;
-      bool ok = p0_ok && opix < M;
+      int opix = opix_b + j;
+      bool ok = p0_ok && opix < Mi;
       if (!b0_is_b) {
         if (ok) {
           *reinterpret_cast<uint4*>(v[j]) =
-              *reinterpret_cast<const uint4*>(p0_gy + opix * Cout);
+              *reinterpret_cast<const uint4*>(p0_gy + (int64_t)opix * Cout);
         } else {
           *reinterpret_cast<uint4*>(v[j]) = uint4{0, 0, 0, 0};
         }
@@ -606,16 +639,16 @@ __global__ __launch_bounds__(256) void conv_wrw3_kernel(
         if (ok && hi >= 0 && hi < Hi && wi >= 0 && wi < Wi) {
           *reinterpret_cast<uint4*>(v[j]) =
               *reinterpret_cast<const uint4*>(
-                  x + (((int64_t)n_ * Hi + hi) * Wi + wi) * Cin + p0_ci);
+                  x + ((int64_t)(n_ * Hi + hi) * Wi + wi) * Cin + p0_ci);
         } else {
           *reinterpret_cast<uint4*>(v[j]) = uint4{0, 0, 0, 0};
         }
-      }
-      if (++wo_ == Wo) {
-        wo_ = 0;
-        if (++ho_ == Ho) {
-          ho_ = 0;
-          ++n_;
+        if (++wo_ == Wo) {
+          wo_ = 0;
+          if (++ho_ == Ho) {
+            ho_ = 0;
+            ++n_;
+          }
         }
       }
     }
@@ -633,24 +666,21 @@ __global__ __launch_bounds__(256) void conv_wrw3_kernel(
     }
   };
 
-  // overflow slot: load + write 4 opix at a time, transient registers
-  auto stage1_half = [&](int kt, int half) {
-    __hip_bfloat16 tmp[4][8];
-    int64_t base_opix = (int64_t)kt * BK + p1_kg * 8 + half * 4;
-    int wo_ = (int)(base_opix % Wo);
-    int64_t r2 = base_opix / Wo;
-    int ho_ = (int)(r2 % Ho);
-    int n_ = (int)(r2 / Ho);
+  // overflow B slot (threads 0..63): load + write 8 opix with transient
+  // registers, one pass
+  auto stage1 = [&](int opix_b, const Coord& cb) {
+    __hip_bfloat16 tmp[8][8];
+    int wo_ = cb.wo, ho_ = cb.ho, n_ = cb.n;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      int64_t opix = base_opix + j;
+    for (int j = 0; j < 8; ++j) {
+      int opix = opix_b + j;
       int hi = ho_ * stride - pad + p1_dh;
       int wi = wo_ * stride - pad + p1_dw;
-      if (p1_ok && opix < M && hi >= 0 && hi < Hi && wi >= 0 &&
+      if (p1_ok && opix < Mi && hi >= 0 && hi < Hi && wi >= 0 &&
           wi < Wi) {
         *reinterpret_cast<uint4*>(tmp[j]) =
             *reinterpret_cast<const uint4*>(
-                x + (((int64_t)n_ * Hi + hi) * Wi + wi) * Cin + p1_ci);
+                x + ((int64_t)(n_ * Hi + hi) * Wi + wi) * Cin + p1_ci);
       } else {
         *reinterpret_cast<uint4*>(tmp[j]) = uint4{0, 0, 0, 0};
       }
@@ -664,27 +694,28 @@ __global__ __launch_bounds__(256) void conv_wrw3_kernel(
     }
 #pragma unroll
     for (int c = 0; c < 8; ++c) {
-      __hip_bfloat16 r[4];
+      __hip_bfloat16 r[8];
 #pragma unroll
-      for (int j = 0; j < 4; ++j) r[j] = tmp[j][c];
-      *reinterpret_cast<uint2*>(
-          sB + lds_byte(p1_row0 + c, p1_kg * 8 + half * 4)) =
-          *reinterpret_cast<const uint2*>(r);
+      for (int j = 0; j < 8; ++j) r[j] = tmp[j][c];
+      *reinterpret_cast<uint4*>(
+          sB + lds_byte(p1_row0 + c, p1_kg * 8)) =
+          *reinterpret_cast<const uint4*>(r);
     }
   };
 
   f32x4 acc[MREP][NREP] = {};
-  if (kt0 < kt1) load_primary(kt0);
+  if (kt0 < kt1) load_primary(p0_base, c0);
   for (int t = kt0; t < kt1; ++t) {
     write_primary();
-    if (has1) {
-      // overflow slot (threads 0..63): loaded + written inline in two
-      // 4-opix halves so its registers stay short-lived and small
-      stage1_half(t, 0);
-      stage1_half(t, 1);
-    }
+    if (has1) stage1(p1_base, c1);
     __syncthreads();
-    if (t + 1 < kt1) load_primary(t + 1);  // loads fly under MFMA
+    if (t + 1 < kt1) {
+      p0_base += BK;
+      coord_step(c0, BK);
+      p1_base += BK;
+      coord_step(c1, BK);
+      load_primary(p0_base, c0);  // loads fly under MFMA
+    }
     {
       // B fragments are read one at a time inside the ni loop so only
       // one is live at once (holding all 9 costs 36 VGPRs and drops
