@@ -68,15 +68,21 @@ def main():
 
         t_ff = timeit(fused_fwd)
         t_fb = timeit(fused_bwd)
-        t_mf = timeit(miopen_fwd)
-        t_mb = timeit(miopen_bwd)
+        skip_ref = os.environ.get("BN_SKIP_MIOPEN") == "1"
+        t_mf = 0.0 if skip_ref else timeit(miopen_fwd)
+        t_mb = 0.0 if skip_ref else timeit(miopen_bwd)
+        # device copy-bandwidth reference at this tensor size: the
+        # streaming ceiling the apply passes should approach
+        t_cp = timeit(lambda: dy.copy_(x))
 
         nbytes = N * C * HW * HW * 2
         # fwd: read x,res write y (+ reduce read x) = 4 passes
         gbs_ff = 4 * nbytes / t_ff / 1e3
         # bwd: reduce reads x,y,dy; apply reads x,y,dy writes dx,dres = 8
         gbs_fb = 8 * nbytes / t_fb / 1e3
+        cp_GBs = 2 * nbytes / t_cp / 1e3
         row = dict(C=C, HW=HW, MB=round(nbytes / 1e6, 1),
+                   copy_GBs=round(cp_GBs),
                    fused_fwd_us=round(t_ff, 1), miopen_fwd_us=round(t_mf, 1),
                    fused_bwd_us=round(t_fb, 1), miopen_bwd_us=round(t_mb, 1),
                    fwd_GBs=round(gbs_ff), bwd_GBs=round(gbs_fb))

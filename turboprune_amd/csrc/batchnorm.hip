@@ -483,22 +483,31 @@ static BNShape bn_shape(const at::Tensor& x) {
   return {x.size(0) * x.size(2) * x.size(3), (int)x.size(1)};
 }
 
+// grid-shape knobs (A/B sweep on hardware; defaults = r01 tuning)
+static int bn_env(const char* name, int dflt) {
+  const char* e = getenv(name);
+  return e && e[0] ? atoi(e) : dflt;
+}
+
 template <typename T>
 static dim3 reduce_grid(int64_t rows, int C) {
+  static int rpt = bn_env("TURBOPRUNE_BN_ROWS", 16);
+  static int cap = bn_env("TURBOPRUNE_BN_RBCAP", 4096);
   int cb = (C / Octet<T>::kN + 7) / 8;   // channel blocks (8 octets each)
-  // ~16 rows per thread; cap total blocks near 4k for >>256-CU fill
-  int64_t rb_want = (rows + 32 * 16 - 1) / (32 * 16);
+  int64_t rb_want = (rows + 32 * rpt - 1) / (32 * rpt);
   int rb = (int)std::min<int64_t>(std::max<int64_t>(rb_want, 1),
-                                  std::max<int64_t>(4096 / cb, 8));
+                                  std::max<int64_t>(cap / cb, 8));
   return dim3(rb, cb);
 }
 
 static int apply_grid(int64_t rows, int C, int vn) {
-  // ~2 octets per thread, up to 8192 blocks: full-chip TLP to hide HBM
-  // latency (2048-block cap left these passes latency-bound)
+  // octets per thread + block cap: full-chip TLP to hide HBM latency
+  // (a 2048-block cap left these passes latency-bound in r01)
+  static int opt = bn_env("TURBOPRUNE_BN_OCT", 2);
+  static int cap = bn_env("TURBOPRUNE_BN_BCAP", 8192);
   int64_t total = rows * (C / vn);
-  int64_t blocks = (total + kBlock * 2 - 1) / (kBlock * 2);
-  return (int)std::min<int64_t>(std::max<int64_t>(blocks, 1), 8192);
+  int64_t blocks = (total + kBlock * opt - 1) / (kBlock * opt);
+  return (int)std::min<int64_t>(std::max<int64_t>(blocks, 1), cap);
 }
 
 bool bn_fast_path_ok(const at::Tensor& x) {
