@@ -433,3 +433,17 @@ def test_conv2d_implicit_wrw_matches_autograd(ext, shape):
     scale = ref.float().abs().max().item()
     assert (got.float() - ref.float()).abs().max().item() < \
         0.05 * max(scale, 1.0)
+
+
+def test_synflow_linearize_restore(ext):
+    """K10: fused sign/abs linearize + sign restore vs torch oracle."""
+    from turboprune_amd.ops import functional as TF
+    torch.manual_seed(5)
+    t = torch.randn(1000003, device=DEV)
+    t[::97] = 0.0
+    orig = t.clone()
+    sign = TF.synflow_linearize_(t)
+    assert sign.dtype == torch.int8
+    assert torch.equal(t, orig.abs())
+    TF.synflow_restore_(t, sign)
+    assert torch.equal(t, orig)

@@ -7,6 +7,8 @@ void mask_from_threshold_(at::Tensor, const at::Tensor&, double);
 at::Tensor masked_abs_score(const at::Tensor&, const at::Tensor&,
                             const at::Tensor&);
 void bernoulli_mask_(at::Tensor, double, int64_t);
+at::Tensor sign_abs_(at::Tensor);
+void mul_sign_(at::Tensor, const at::Tensor&);
 void sgd_step_(at::Tensor, const at::Tensor&, at::Tensor, const at::Tensor&,
                at::Tensor, double, double, double);
 double kth_smallest(const at::Tensor&, int64_t);
@@ -90,6 +92,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "|w*m| or |w*m*g|");
   m.def("bernoulli_mask_", &turboprune::bernoulli_mask_,
         "Philox Bernoulli(p) mask fill");
+  m.def("sign_abs_", &turboprune::sign_abs_,
+        "SynFlow linearize: t = |t| in place, returns int8 signs (K10)");
+  m.def("mul_sign_", &turboprune::mul_sign_,
+        "SynFlow restore: t *= sign");
   m.def("sgd_step_", &turboprune::sgd_step_,
         "fused SGD momentum+wd step with mask-reapply cache rewrite");
   m.def("sgd_step_multi_", &turboprune::sgd_step_multi_,

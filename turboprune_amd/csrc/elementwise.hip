@@ -304,4 +304,52 @@ void colsum2_atomic(const at::Tensor& pa, const at::Tensor& pb,
                      ob.data_ptr<float>(), R, C);
 }
 
+// ---- SynFlow linearize / restore ---------------------------
+// One fused pass each way: linearize emits the sign as int8 (4x smaller
+// than the reference's fp32 sign tensors, pruning_utils.py:223-248) and
+// takes |t| in place; restore multiplies the sign back.
+__global__ void sign_abs_kernel(float* __restrict__ t,
+                                int8_t* __restrict__ sign, int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float v = t[i];
+    sign[i] = v > 0.f ? 1 : (v < 0.f ? -1 : 0);
+    t[i] = fabsf(v);
+  }
+}
+
+__global__ void mul_sign_kernel(float* __restrict__ t,
+                                const int8_t* __restrict__ sign,
+                                int64_t n) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    t[i] *= (float)sign[i];
+}
+
+at::Tensor sign_abs_(at::Tensor t) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kFloat &&
+              t.is_non_overlapping_and_dense(),
+              "sign_abs_: dense fp32 GPU tensor expected");
+  auto sign = at::empty_like(t, t.options().dtype(at::kChar));
+  int64_t n = t.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(sign_abs_kernel, dim3(elementwise_grid(n)),
+                     dim3(kBlock), 0, stream, t.data_ptr<float>(),
+                     sign.data_ptr<int8_t>(), n);
+  return sign;
+}
+
+void mul_sign_(at::Tensor t, const at::Tensor& sign) {
+  TORCH_CHECK(t.is_cuda() && t.scalar_type() == at::kFloat &&
+              sign.scalar_type() == at::kChar &&
+              t.numel() == sign.numel());
+  int64_t n = t.numel();
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(mul_sign_kernel, dim3(elementwise_grid(n)),
+                     dim3(kBlock), 0, stream, t.data_ptr<float>(),
+                     sign.data_ptr<int8_t>(), n);
+}
+
 }  // namespace turboprune

@@ -133,6 +133,27 @@ class _MaskedLinearGemm(torch.autograd.Function):
         return grad_x, grad_w, grad_b
 
 
+def synflow_linearize_(t: torch.Tensor):
+    """t <- |t| in place; returns the sign tensor (reference
+    pruning_utils.py:223-248). GPU: one fused HIP pass emitting int8
+    signs; CPU oracle: torch sign + abs_."""
+    if _backend.use_native(t) and t.dtype == torch.float32 \
+            and t.is_contiguous():
+        return _backend.extension().sign_abs_(t)
+    s = torch.sign(t)
+    t.abs_()
+    return s
+
+
+def synflow_restore_(t: torch.Tensor, sign: torch.Tensor):
+    """t <- t * sign in place (signs from synflow_linearize_)."""
+    if _backend.use_native(t) and t.dtype == torch.float32 \
+            and sign.dtype == torch.int8 and t.is_contiguous():
+        _backend.extension().mul_sign_(t, sign)
+        return t
+    return t.mul_(sign.to(t.dtype))
+
+
 def bernoulli_mask_(mask: torch.Tensor, p: float,
                     seed: Optional[int] = None) -> torch.Tensor:
     """In-place Bernoulli(p) fill of a mask buffer (reference:

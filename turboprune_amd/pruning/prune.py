@@ -109,13 +109,13 @@ def prune_synflow(model: nn.Module, density: float,
     device = device or _model_device(model)
 
     # linearize: theta <- |theta| over ALL params & buffers, keep signs
+    # (fused HIP pass on GPU — K10)
     signs = {}
     with torch.no_grad():
         state = model.state_dict()
         for name, t in state.items():
             if t.is_floating_point():
-                signs[name] = torch.sign(t)
-                t.abs_()
+                signs[name] = TF.synflow_linearize_(t)
 
     inputs, _ = next(iter(dataloader))
     input_shape = list(inputs.shape)
@@ -142,7 +142,7 @@ def prune_synflow(model: nn.Module, density: float,
         state = model.state_dict()
         for name, t in state.items():
             if name in signs:
-                t.mul_(signs[name])
+                TF.synflow_restore_(t, signs[name])
         thr = _threshold_global(list(scores.values()), density)
         if thr is not None:
             _rewrite_masks(model, scores, thr)
