@@ -504,7 +504,7 @@ static int apply_grid(int64_t rows, int C, int vn) {
   // octets per thread + block cap: full-chip TLP to hide HBM latency
   // (a 2048-block cap left these passes latency-bound in r01)
   static int opt = bn_env("TURBOPRUNE_BN_OCT", 2);
-  static int cap = bn_env("TURBOPRUNE_BN_BCAP", 8192);
+  static int cap = bn_env("TURBOPRUNE_BN_BCAP", 16384);
   int64_t total = rows * (C / vn);
   int64_t blocks = (total + kBlock * opt - 1) / (kBlock * opt);
   return (int)std::min<int64_t>(std::max<int64_t>(blocks, 1), cap);
