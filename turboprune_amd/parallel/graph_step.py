@@ -115,7 +115,10 @@ class GraphedTrainStep:
 
 
 def wanted(per_rank_batch: int, distributed: bool) -> bool:
-    """Auto-enable heuristic: replay pays when the step is launch-bound
-    (small per-rank batch, the strong-scaling regime); at bs >= 256 the
-    kernels are long enough that DDP's backward-overlapped buckets win."""
-    return per_rank_batch <= 128
+    """Auto-enable heuristic — currently OFF by measurement: at bs64 on
+    MI355X the captured replay matched eager within noise (16.9 vs
+    16.5 ms, r2d) because the small-batch step is GPU-busy, not
+    launch-bound, and the captured flat all-reduce gives up DDP's
+    backward overlap. Capture stays available via --graph; flip this
+    back if a future kernel generation makes the step launch-bound."""
+    return False
