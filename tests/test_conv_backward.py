@@ -63,3 +63,42 @@ def test_grad_input_fuzz_geometries():
         y.backward(gy)
         gx = conv_grad_input(gy, w, (hi, hi), s, pad)
         torch.testing.assert_close(gx, x.grad, rtol=1e-4, atol=1e-4)
+
+
+@pytest.mark.parametrize("shape", [
+    # every ResNet50 stride-2 geometry (cin, cout, k, hi) at p=k//2
+    (64, 128, 3, 16),
+    (128, 128, 3, 56),
+    (64, 256, 1, 56),
+    (128, 512, 1, 28),
+    (32, 64, 3, 15),   # odd spatial
+])
+def test_conv_grad_input_s2_parity_oracle(shape):
+    """Parity-class stride-2 gradin == autograd's grad (fp32 CPU)."""
+    from turboprune_amd.ops.conv_backward import conv_grad_input_s2_parity
+    cin, cout, k, hi = shape
+    p = k // 2
+    torch.manual_seed(cin + k)
+    x = torch.randn(2, cin, hi, hi, requires_grad=True)
+    w = torch.randn(cout, cin, k, k, requires_grad=True) * 0.1
+    y = torch.nn.functional.conv2d(x, w, None, 2, p)
+    gy = torch.randn_like(y)
+    (ref,) = torch.autograd.grad(y, x, gy)
+    got = conv_grad_input_s2_parity(gy, w.detach(), (hi, hi), p)
+    assert torch.allclose(got, ref, atol=1e-4), \
+        (got - ref).abs().max().item()
+
+
+def test_conv_grad_input_s2_parity_channels_last():
+    from turboprune_amd.ops.conv_backward import conv_grad_input_s2_parity
+    torch.manual_seed(0)
+    x = torch.randn(2, 8, 12, 12, requires_grad=True)
+    w = torch.randn(16, 8, 3, 3) * 0.1
+    y = torch.nn.functional.conv2d(x, w, None, 2, 1)
+    gy = torch.randn_like(y).contiguous(
+        memory_format=torch.channels_last)
+    (ref,) = torch.autograd.grad(y, x, gy)
+    got = conv_grad_input_s2_parity(
+        gy, w.contiguous(memory_format=torch.channels_last), (12, 12), 1)
+    assert got.is_contiguous(memory_format=torch.channels_last)
+    assert torch.allclose(got, ref, atol=1e-4)

@@ -60,3 +60,61 @@ def conv_grad_input(gy: torch.Tensor, w: torch.Tensor,
     w_rt = rot180_transpose(w)
     gy_d = dilate_gy(gy, stride, in_hw, k, pad)
     return conv_fn(gy_d, w_rt, 1, k - 1 - pad)
+
+
+def conv_grad_input_s2_parity(gy: torch.Tensor, w: torch.Tensor,
+                              in_hw: Tuple[int, int], pad: int,
+                              conv_fn: Optional[Callable] = None
+                              ) -> torch.Tensor:
+    """Stride-2 grad_input as FOUR dense stride-1 sub-convolutions
+    (output-parity classes) instead of the zero-dilated conv (which
+    spends 4x the MFMA work multiplying inserted zeros — measured 3-4x
+    slower than MIOpen, profiles/r02_conv_dispatch.md).
+
+    For k=3,s=2,p=1 (and the k=1,s=2,p=0 degenerate case): an output
+    pixel gx[2a+rh, 2b+rw] only receives taps with dh ≡ (rh+p) mod 2,
+    so each parity class (rh, rw) is a stride-1 conv of gy with a
+    (rh+1)x(rw+1) sub-kernel of the rotated-transposed weight; results
+    interleave back by strided scatter. Total FLOP = the exact gradin
+    FLOP, zero dilation waste."""
+    conv_fn = conv_fn or (lambda a, b, s, p: F.conv2d(a, b, None, s, p))
+    k = w.shape[2]
+    hi, wi = in_hw
+    n, cout, ho, wo = gy.shape
+    cin = w.shape[1]
+    cl = gy.is_contiguous(memory_format=torch.channels_last)
+    wf = rot180_transpose(w)  # (Cin, Cout, k, k), taps flipped
+
+    if k == 1:
+        # only the (p, p)-parity class exists: plain 1x1 stride-1 conv
+        # scattered to every stride-th position, zeros elsewhere
+        assert pad == 0
+        out = gy.new_zeros(n, cin, hi, wi)
+        if cl:
+            out = out.contiguous(memory_format=torch.channels_last)
+        cls = conv_fn(gy, wf, 1, 0)
+        out[:, :, ::2, ::2] = cls[:, :, :(hi + 1) // 2, :(wi + 1) // 2]
+        return out
+
+    assert k == 3 and pad == 1, "parity path covers k3p1/k1p0 stride-2"
+    # flipped-tap index sets per parity (ascending gy-offset order):
+    # rh=0 -> original dh=1 -> flipped index 1; rh=1 -> original {2,0}
+    # -> flipped {0,2}
+    sel = {0: [1], 1: [0, 2]}
+    # one bottom/right-padded gy so every class runs pad-free
+    gy_p = F.pad(gy, (0, 1, 0, 1))
+    if cl:
+        gy_p = gy_p.contiguous(memory_format=torch.channels_last)
+    out = torch.empty(n, cin, hi, wi, dtype=gy.dtype, device=gy.device)
+    if cl:
+        out = out.contiguous(memory_format=torch.channels_last)
+    for rh in (0, 1):
+        for rw in (0, 1):
+            sub = wf[:, :, sel[rh]][:, :, :, sel[rw]]
+            sub = sub.contiguous(memory_format=torch.channels_last) \
+                if cl else sub.contiguous()
+            cls = conv_fn(gy_p, sub, 1, 0)
+            nh = (hi - rh + 1) // 2  # rows of this parity class in gx
+            nw = (wi - rw + 1) // 2
+            out[:, :, rh::2, rw::2] = cls[:, :, :nh, :nw]
+    return out

@@ -54,9 +54,20 @@ class NativeBackend:
 
     @staticmethod
     def gradin(gy, w, in_hw, stride, pad):
-        # one fused kernel: rotated-weight forward conv with the
-        # zero-inserted gy coordinates resolved inside the im2col gather
         from turboprune_amd.ops._backend import extension
+        if stride == 2 and os.environ.get("TURBOPRUNE_GRADIN_S2",
+                                          "parity") == "parity":
+            # four dense stride-1 sub-convs (zero dilation waste) on
+            # the implicit-GEMM fwd kernel; see conv_backward.py
+            from turboprune_amd.ops.conv_backward import \
+                conv_grad_input_s2_parity
+            ext = extension()
+            return conv_grad_input_s2_parity(
+                gy, w, in_hw, pad,
+                conv_fn=lambda a, b, s, p:
+                ext.conv2d_implicit_fwd(a, b, None, s, p))
+        # fused kernel: rotated-weight forward conv with the
+        # zero-inserted gy coordinates resolved inside the im2col gather
         return extension().conv2d_implicit_gradin(
             gy, w, in_hw[0], in_hw[1], stride, pad)
 
