@@ -95,8 +95,9 @@ def main():
         t_l = timeit(lambda: F.conv2d(x, w, None, s, p))
         row["fwd"] = {"native_us": round(t_n, 1), "lib_us": round(t_l, 1)}
 
-        # grad_input (native fused dilated gather for stride 2)
-        t_n = timeit(lambda: ext.conv2d_implicit_gradin(gy, w, hi, hi, s, p))
+        # grad_input through NativeBackend (parity sub-convs for s2)
+        from turboprune_amd.ops.conv_native import NativeBackend
+        t_n = timeit(lambda: NativeBackend.gradin(gy, w, (hi, hi), s, p))
         t_l = timeit(lambda: torch.ops.aten.convolution_backward(
             gy, x, w, None, (s, s), (p, p), (1, 1), False, (0, 0), 1,
             (True, False, False))[0])
