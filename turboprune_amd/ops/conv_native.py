@@ -56,9 +56,12 @@ class NativeBackend:
     def gradin(gy, w, in_hw, stride, pad):
         from turboprune_amd.ops._backend import extension
         if stride == 2 and os.environ.get("TURBOPRUNE_GRADIN_S2",
-                                          "parity") == "parity":
-            # four dense stride-1 sub-convs (zero dilation waste) on
-            # the implicit-GEMM fwd kernel; see conv_backward.py
+                                          "dilated") == "parity":
+            # four dense stride-1 sub-convs; measured r2j: the eager
+            # pad + strided-scatter glue makes this SLOWER than the
+            # fused dilated kernel on device (1.0-1.7 ms vs 0.8-0.9),
+            # so it stays opt-in until the scatter is fused into the
+            # conv epilogue; see conv_backward.py
             from turboprune_amd.ops.conv_backward import \
                 conv_grad_input_s2_parity
             ext = extension()
