@@ -12,10 +12,11 @@ import argparse
 import json
 import sys
 
-# (model, global_batch, n_gpus) -> images/sec floor (round-1 measured)
+# (model, global_batch, n_gpus) -> images/sec floor (round-2 measured:
+# auto conv dispatch 8151-8175, DeiT with per-shape GEMM routing 7239)
 FLOORS = {
-    ("resnet50", 512, 1): 8015.0,
-    ("deit_small", 256, 1): 7178.0,
+    ("resnet50", 512, 1): 8150.0,
+    ("deit_small", 256, 1): 7239.0,
 }
 
 
