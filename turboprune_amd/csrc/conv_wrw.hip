@@ -754,6 +754,245 @@ __global__ __launch_bounds__(256) void conv_wrw3_kernel(
   }
 }
 
+// ---- v4: glds raw staging + in-LDS transpose ----------------------------
+//
+// What v2/v3 established (profiles/r02_summary.md): a winning wrw must
+// simultaneously (a) keep operand re-reads ~1x (v3's 64x576 tile),
+// (b) keep >=10 KB of global loads in flight per CU (v3's
+// register-held gathers managed 8x16 B per wave at occupancy 1:
+// ~0.9 TB/s), (c) spill nothing, (d) spend ~0 registers on staging.
+// The only structure that satisfies all four is the fwd kernel's:
+// `global_load_lds` — per-lane gathered addresses, zero destination
+// registers, fire-and-forget depth limited only by vmcnt. glds cannot
+// transpose, so the tile lands in LDS in gather order ([opix][c]) and
+// a separate LDS->LDS transpose stage builds the [row][k] MFMA images.
+//
+// Per 64co x 576tapci x 32opix tile:
+//   glds:  A raw 4 KB in 4 instrs + B raw 36 KB in 36 instrs
+//          (each instr: 64 lanes x 16 B -> 1 KB contiguous deposit;
+//          per-lane bounds select to a zero page, like conv_implicit)
+//   MFMA(t) runs while tile t+1's glds are in flight
+//   transpose: each thread moves ~1.25 row-octets: 8 strided
+//          ds_read_b128 from raw + register transpose + 8 ds_write_b128
+//          into the swizzled image (short-lived registers)
+// LDS: rawA 4K + rawB 36K + imgA 4K + imgB 36K = 80 KiB dynamic
+// -> exactly 2 blocks/CU.
+namespace conv_wrw4 {
+constexpr int BM = 64, BN = 576, BK = 32;
+constexpr int WN = 144;
+constexpr int MREP = 4, NREP = 9;
+constexpr int RAW_A = 0;                  // [32 opix][64 co]   4 KiB
+constexpr int RAW_B = 4 * 1024;           // [9 chunk][32 opix][64 ci]
+constexpr int IMG_A = RAW_A + 40 * 1024;  // swizzled [64 co][32 k]
+constexpr int IMG_B = IMG_A + 4 * 1024;   // swizzled [576][32 k]
+TP_DEVICE int lds_byte(int row, int k) {  // same image as v3
+  int blk = ((k >> 3) ^ row) & 3;
+  return row * (BK * 2) + blk * 16 + (k & 7) * 2;
+}
+}  // namespace conv_wrw4
+
+__global__ __launch_bounds__(256) void conv_wrw4_kernel(
+    const __hip_bfloat16* __restrict__ gy,  // (M, Cout) row-major
+    const __hip_bfloat16* __restrict__ x,   // NHWC
+    const __hip_bfloat16* __restrict__ zero_page,  // >=128 B of zeros
+    float* __restrict__ partial,            // (slabs, Mp, Np)
+    int Nb, int Hi, int Wi, int Cin, int Cout, int Ho, int Wo, int KH,
+    int KW, int stride, int pad, int Mp, int Np, int grid_n) {
+  using namespace conv_wrw4;
+  extern __shared__ char smem[];
+
+  int wg = blockIdx.x;
+  int tile_m = (wg / grid_n) * BM;  // co tile
+  int tile_n = (wg % grid_n) * BN;  // tapci tile
+  int64_t M64 = (int64_t)Nb * Ho * Wo;
+  int Mi = (int)M64;
+
+  int lane = threadIdx.x & (kWave - 1);
+  int wid = threadIdx.x / kWave;
+
+  int total_kt = (int)((M64 + BK - 1) / BK);
+  int per = (total_kt + gridDim.y - 1) / gridDim.y;
+  int kt0 = blockIdx.y * per;
+  int kt1 = min(kt0 + per, total_kt);
+  partial += (int64_t)blockIdx.y * Mp * Np;
+
+  // ---- glds staging ----------------------------------------------------
+  // wave w issues: 1 A-instr (opix rows 8w..8w+7) + 9 B-instrs
+  // (chunk c = w + 4*i, og cycling). Per-lane opix for an instr with
+  // opix-group og: og*8 + (lane>>3). Each lane tracks FOUR coordinate
+  // sets (its opix offset + {0,8,16,24}), advanced by +BK per tile with
+  // carry loops — no divisions in the loop.
+  int lrow = lane >> 3;   // 0..7: opix within group
+  int loct = lane & 7;    // 16 B octet within the 128 B segment
+
+  int Ktot = KH * KW * Cin;
+  // B chunk geometry (9 chunks of 64 tapci): tap + ci0 per chunk
+  // (Cin % 64 == 0 keeps each chunk inside one tap)
+  int cdh[9], cdw[9], cci[9];
+  bool cok[9];
+#pragma unroll
+  for (int c = 0; c < 9; ++c) {
+    int tapci = tile_n + c * 64;
+    cok[c] = tapci < Ktot;
+    int tap = cok[c] ? tapci / Cin : 0;
+    cci[c] = cok[c] ? tapci % Cin : 0;
+    cdh[c] = tap / KW;
+    cdw[c] = tap % KW;
+  }
+
+  // per-lane coordinate sets for opix = kt*BK + og*8 + lrow, og=0..3
+  int wo4[4], ho4[4], n4[4], op4[4];
+#pragma unroll
+  for (int og = 0; og < 4; ++og) {
+    int opix = kt0 * BK + og * 8 + lrow;
+    op4[og] = opix;
+    int o = opix < Mi ? opix : 0;
+    wo4[og] = o % Wo;
+    int r2 = o / Wo;
+    ho4[og] = r2 % Ho;
+    n4[og] = r2 / Ho;
+  }
+  auto advance = [&]() {
+#pragma unroll
+    for (int og = 0; og < 4; ++og) {
+      op4[og] += BK;
+      wo4[og] += BK;
+      while (wo4[og] >= Wo) {
+        wo4[og] -= Wo;
+        if (++ho4[og] == Ho) {
+          ho4[og] = 0;
+          ++n4[og];
+        }
+      }
+    }
+  };
+
+  auto issue_glds = [&]() {
+    // A: this wave's 8 opix rows (og = wid & 3 happens to be wid)
+    {
+      int og = wid;
+      const char* src;
+      if (tile_m + loct * 8 < Cout && op4[og] < Mi) {
+        src = reinterpret_cast<const char*>(
+            gy + (int64_t)op4[og] * Cout + tile_m + loct * 8);
+      } else {
+        src = reinterpret_cast<const char*>(zero_page) + loct * 16;
+      }
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(
+              smem + RAW_A + og * 8 * 128),
+          16, 0, 0);
+    }
+    // B: 9 instrs per wave: (chunk, og) pairs; c fixed per instr,
+    // og = instr index & 3 keeps all four coord sets busy
+#pragma unroll
+    for (int i = 0; i < 9; ++i) {
+      int c = i;           // chunk
+      int og = (i + wid) & 3;
+      int hi = ho4[og] * stride - pad + cdh[c];
+      int wi = wo4[og] * stride - pad + cdw[c];
+      bool ok = cok[c] && op4[og] < Mi && hi >= 0 && hi < Hi &&
+                wi >= 0 && wi < Wi;
+      const char* src =
+          ok ? reinterpret_cast<const char*>(
+                   x + ((int64_t)(n4[og] * Hi + hi) * Wi + wi) * Cin +
+                   cci[c] + loct * 8)
+             : reinterpret_cast<const char*>(zero_page) + loct * 16;
+      // each WAVE must deposit a distinct 1 KB run; runs are indexed
+      // by (chunk, og): raw_b[chunk][og*8 + lrow][loct]
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(
+              smem + RAW_B + (c * 32 + og * 8) * 128),
+          16, 0, 0);
+    }
+  };
+
+  // ---- transpose raw -> swizzled images --------------------------------
+  // slots: 0..31 A (co-octet ro=s>>2, kg=s&3), 32..319 B (tapci-octet).
+  // Threads 0..63 take a second slot.
+  int s0 = threadIdx.x;
+  int s1 = threadIdx.x + 256;
+  bool has1 = threadIdx.x < 64;
+
+  auto do_slot = [&](int s) {
+    bool is_b = s >= 32;
+    int ro = is_b ? (s - 32) >> 2 : s >> 2;  // row-octet
+    int kg = is_b ? (s - 32) & 3 : s & 3;    // 8-opix group
+    // raw read base: A raw [opix][co]: byte = opix*128 + ro*16
+    //                B raw chunk = ro>>3, ci-octet = ro&7
+    int raw_base = is_b ? RAW_B + (ro >> 3) * 4096 + (ro & 7) * 16
+                        : RAW_A + ro * 16;
+    char* img = smem + (is_b ? IMG_B : IMG_A);
+    __hip_bfloat16 t8[8][8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      *reinterpret_cast<uint4*>(t8[j]) = *reinterpret_cast<const uint4*>(
+          smem + raw_base + (kg * 8 + j) * 128);
+    }
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      __hip_bfloat16 r[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) r[j] = t8[j][c];
+      *reinterpret_cast<uint4*>(img + lds_byte(ro * 8 + c, kg * 8)) =
+          *reinterpret_cast<const uint4*>(r);
+    }
+  };
+
+  f32x4 acc[MREP][NREP] = {};
+  if (kt0 < kt1) {
+    issue_glds();   // tile kt0 in flight
+    __syncthreads();  // carries vmcnt(0): deposits landed (guide §5.4)
+  }
+  for (int t = kt0; t < kt1; ++t) {
+    // raw(t) is complete (drained by the previous barrier): transpose
+    do_slot(s0);
+    if (has1) do_slot(s1);
+    __syncthreads();  // images ready; raw free
+    // issue tile t+1's gathers — they fly under the MFMAs and are
+    // drained by the loop-end barrier
+    if (t + 1 < kt1) {
+      advance();
+      issue_glds();
+    }
+    {
+      bf16x8 a_frag[MREP];
+      int kf = (lane >> 4) * 8;
+      int rowf = lane & 15;
+#pragma unroll
+      for (int mi = 0; mi < MREP; ++mi)
+        a_frag[mi] = *reinterpret_cast<const bf16x8*>(
+            smem + IMG_A + lds_byte(mi * 16 + rowf, kf));
+#pragma unroll
+      for (int ni = 0; ni < NREP; ++ni) {
+        bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+            smem + IMG_B + lds_byte(wid * WN + ni * 16 + rowf, kf));
+#pragma unroll
+        for (int mi = 0; mi < MREP; ++mi)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag, acc[mi][ni], 0, 0, 0);
+      }
+    }
+    // images consumed + next tile's glds drained before re-transpose
+    __syncthreads();
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MREP; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < NREP; ++ni) {
+      int col = tile_n + wid * WN + ni * 16 + (lane & 15);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = tile_m + mi * 16 + (lane >> 4) * 4 + j;
+        partial[(int64_t)row * Np + col] = acc[mi][ni][j];
+      }
+    }
+  }
+}
+
 // fp32 output: the split-K accumulation is fp32 and the consumer is the
 // fp32 master-weight gradient — rounding to bf16 here would be a
 // systematic numerics divergence vs the reference autocast path
@@ -795,18 +1034,68 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
   int grid_m = Mp / BM, grid_n = Np / BN;
   int tiles = grid_m * grid_n;
 
-  // kernel version: auto = v3 for 3x3 (wide-N tiles, minimal re-reads),
-  // v2 otherwise; TURBOPRUNE_WRW=2 forces v2, =1 forces v1,
-  // TURBOPRUNE_WRW_DB=1 the v1 double-buffered variant (A/B knobs)
-  static int use_db = -1, use_v1 = -1, force_v2 = -1;
+  // kernel version: auto = v4 for 3x3 (glds staging + in-LDS
+  // transpose), v2 otherwise; TURBOPRUNE_WRW=3 forces v3, =2 v2,
+  // =1 v1, TURBOPRUNE_WRW_DB=1 the v1 double-buffered variant
+  static int use_db = -1, use_v1 = -1, force_v2 = -1, force_v3 = -1;
   if (use_db < 0) {
     const char* e = getenv("TURBOPRUNE_WRW_DB");
     use_db = (e && e[0] == '1') ? 1 : 0;
     e = getenv("TURBOPRUNE_WRW");
     use_v1 = (e && e[0] == '1') ? 1 : 0;
     force_v2 = (e && e[0] == '2') ? 1 : 0;
+    force_v3 = (e && e[0] == '3') ? 1 : 0;
   }
   auto stream = at::hip::getCurrentHIPStream();
+
+  if (!use_db && !use_v1 && !force_v2 && !force_v3 && KH == 3 &&
+      KW == 3) {
+    // v4: same tiling/split-K as v3, glds-staged
+    constexpr int BM4 = conv_wrw4::BM, BN4 = conv_wrw4::BN,
+                  BK4 = conv_wrw4::BK;
+    int gm = (Cout + BM4 - 1) / BM4;
+    int gn = (K + BN4 - 1) / BN4;
+    int Mp4 = gm * BM4;
+    int Np4 = gn * BN4;
+    int tiles4 = gm * gn;
+    int total_kt = (int)((M + BK4 - 1) / BK4);
+    int splitk = 1;
+    while (tiles4 * splitk < 1024 && splitk * 2 <= total_kt &&
+           splitk < 512)
+      splitk *= 2;
+    auto partial = at::empty({splitk, (int64_t)Mp4, (int64_t)Np4},
+                             x.options().dtype(at::kFloat));
+    static at::Tensor zero_page;
+    if (!zero_page.defined() || zero_page.device() != x.device())
+      zero_page = at::zeros({64}, x.options());
+    constexpr int kLds = 80 * 1024;
+    static bool attr_set = false;
+    if (!attr_set) {
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(conv_wrw4_kernel),
+          hipFuncAttributeMaxDynamicSharedMemorySize, kLds);
+      attr_set = true;
+    }
+    hipLaunchKernelGGL(conv_wrw4_kernel, dim3(tiles4, splitk), dim3(256),
+                       kLds, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(gy.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(
+                           zero_page.data_ptr()),
+                       partial.data_ptr<float>(), Nb, Hi, Wi, Cin, Cout,
+                       Ho, Wo, (int)KH, (int)KW, (int)stride, (int)pad,
+                       Mp4, Np4, gn);
+    auto gw = at::empty({Cout, Cin, KH, KW},
+                        gy.options().dtype(at::kFloat).memory_format(
+                            at::MemoryFormat::ChannelsLast));
+    int64_t mn = (int64_t)Cout * K;
+    int rgrid = elementwise_grid(mn, kBlock, 4);
+    hipLaunchKernelGGL(wrw_reduce_kernel, dim3(rgrid), dim3(kBlock), 0,
+                       stream, partial.data_ptr<float>(),
+                       gw.data_ptr<float>(), mn,
+                       (int64_t)Mp4 * Np4, splitk, Np4, K, (int64_t)K);
+    return gw;
+  }
 
   if (!use_db && !use_v1 && !force_v2 && KH == 3 && KW == 3) {
     // v3: 64 x 576 tiles — gy re-read ceil(K/576)x, x re-read
