@@ -1,10 +1,13 @@
 """Scaled-dot-product attention for the DeiT path.
 
-Default is torch SDPA (the validated round-1 configuration). The fused
-flash-style HIP forward (csrc/attention.hip: online softmax, one block
-per (batch*head, 64-q-row tile), head_dim 64) is opt-in via
-``TURBOPRUNE_ATTN=native`` once scripts/validate_attention.py has passed
-on device — it has NOT run on hardware yet.
+Default is torch SDPA, BY MEASUREMENT: the fused flash-style HIP
+forward (csrc/attention.hip: online softmax, one block per
+(batch*head, 64-q-row tile), head_dim 64) is numerically validated on
+device (scripts/validate_attention.py, r2 checks) but runs 182 µs vs
+AOTriton SDPA's 90 µs on the DeiT shape and costs −12% DeiT e2e, so it
+stays opt-in via ``TURBOPRUNE_ATTN=native`` until tuned
+(docs/ROADMAP_ROUND3.md item 3). SURVEY §5 allows SDPA to remain
+torch-ROCm for this workload.
 
 Backward is closed-form recompute (the flash-attention backward
 identities), expressed in torch ops so it works with either forward and
