@@ -993,6 +993,279 @@ __global__ __launch_bounds__(256) void conv_wrw4_kernel(
   }
 }
 
+// ---- v5: persistent ring window (3x3 stride-1, large Wo) ---------------
+//
+// v4's residual wall is gather THROUGHPUT: the im2col gather moves 9x
+// the x bytes through L2 at 16 B granularity (r2l PMC). Observation:
+// at BK=32 opix a tile advances only ~0.5-1 OUTPUT ROW, and for
+// k3/s1/p1 the x rows a tile needs are the CONTIGUOUS flat-row range
+// [R0-1, R1+1] (Hi == Ho, flat row F = n*Hi + hi = R + dh - 1; the
+// cross-image bleed rows are masked at transpose time). So keep a
+// 5-row ring window of x RESIDENT in LDS and, per tile, load only the
+// 0-2 NEW rows — pure contiguous 1 KB glds (7 per row), x read ~once
+// total. The 9-tap resolution happens during the LDS transpose (window
+// reads), which is on-chip. Covers the shapes where the window (5 rows
+// x Wi x Cin x 2 B = 35.8 KiB) fits and rows are 7 x 1 KiB exactly:
+// layer1 (56²,Cin64) and layer2 (28²,Cin128) — the two biggest MIOpen
+// wrw entries. Others fall back to v4.
+namespace conv_wrw5 {
+constexpr int BM = 64, BN = 576, BK = 32;
+constexpr int WN = 144;
+constexpr int MREP = 4, NREP = 9;
+constexpr int WROWS = 5;
+constexpr int RAW_A = 0;                     //  4 KiB
+constexpr int WIN = 4 * 1024;                // 36 KiB (5 x 7168 B ring)
+constexpr int IMG_A = WIN + 36 * 1024;       //  4 KiB
+constexpr int IMG_B = IMG_A + 4 * 1024;      // 36 KiB
+TP_DEVICE int lds_byte(int row, int k) {
+  int blk = ((k >> 3) ^ row) & 3;
+  return row * (BK * 2) + blk * 16 + (k & 7) * 2;
+}
+}  // namespace conv_wrw5
+
+__global__ __launch_bounds__(256) void conv_wrw5_kernel(
+    const __hip_bfloat16* __restrict__ gy,  // (M, Cout) row-major
+    const __hip_bfloat16* __restrict__ x,   // NHWC
+    float* __restrict__ partial,            // (slabs, Mp, Np)
+    int Nb, int HiWi, int Cin, int Cout, int HoWo, int Mp, int Np,
+    int grid_n) {
+  // HiWi == HoWo (k3 s1 p1); square images: Hi=Wi=HoWo? No — HiWi is
+  // the SIDE (Hi == Wi == Ho == Wo for the gated shapes).
+  using namespace conv_wrw5;
+  extern __shared__ char smem[];
+  const int Wo = HoWo, Hi = HiWi, Wi = HiWi, Ho = HoWo;
+  const int rowpitch = Wi * Cin * 2;        // == 7168 by the host gate
+
+  int wg = blockIdx.x;
+  int tile_m = (wg / grid_n) * BM;  // co tile
+  int tile_n = (wg % grid_n) * BN;  // tapci tile
+  int64_t M64 = (int64_t)Nb * Ho * Wo;
+  int Mi = (int)M64;
+  int Frows = Nb * Hi;              // total flat x rows
+
+  int lane = threadIdx.x & (kWave - 1);
+  int wid = threadIdx.x / kWave;
+
+  int total_kt = (int)((M64 + BK - 1) / BK);
+  int per = (total_kt + gridDim.y - 1) / gridDim.y;
+  int kt0 = blockIdx.y * per;
+  int kt1 = min(kt0 + per, total_kt);
+  partial += (int64_t)blockIdx.y * Mp * Np;
+
+  // ---- uniform per-tile scalar state -----------------------------------
+  int opix0 = kt0 * BK;
+  int wo0 = opix0 % Wo;             // one division at entry only
+  int R0 = opix0 / Wo;              // flat output row of the first opix
+  int hoabs0 = R0 % Ho;
+  int m5 = R0 % WROWS;
+  int Floaded = R0 - 2;             // highest flat x row in the ring
+
+  // B chunk geometry (9 chunks of 64 tapci)
+  int Ktot = 9 * Cin;
+  int cdh[9], cdw[9], cci[9];
+  bool cok[9];
+#pragma unroll
+  for (int c = 0; c < 9; ++c) {
+    int tapci = tile_n + c * 64;
+    cok[c] = tapci < Ktot;
+    int tap = cok[c] ? tapci / Cin : 0;
+    cci[c] = cok[c] ? tapci % Cin : 0;
+    cdh[c] = tap / 3;
+    cdw[c] = tap % 3;
+  }
+
+  // A glds (identical to v4): this wave's 8 opix rows
+  int lrow = lane >> 3;
+  int loct = lane & 7;
+  int opA = opix0 + wid * 8 + lrow;   // per-lane A opix
+
+  auto issue_A = [&]() {
+    const char* src;
+    if (opA < Mi) {
+      src = reinterpret_cast<const char*>(
+          gy + (int64_t)opA * Cout + tile_m + loct * 8);
+    } else {
+      src = reinterpret_cast<const char*>(x);  // any valid memory
+    }
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)src,
+        (__attribute__((address_space(3))) void*)(
+            smem + RAW_A + wid * 8 * 128),
+        16, 0, 0);
+  };
+
+  // window loader: bring flat rows (Floaded, target] into the ring.
+  // Row F -> ring slot F mod 5; 7 contiguous 1 KB glds per row,
+  // instr i of the row handled by wave (i % 4).
+  auto load_rows = [&](int target) {
+    for (int F = Floaded + 1; F <= target; ++F) {
+      int Fc = F < 0 ? 0 : (F >= Frows ? Frows - 1 : F);
+      int slot = F % WROWS;
+      if (slot < 0) slot += WROWS;
+      const char* base =
+          reinterpret_cast<const char*>(x) + (int64_t)Fc * rowpitch;
+      char* dst = smem + WIN + slot * rowpitch;
+      int nq = rowpitch >> 10;  // 1 KiB chunks per row (host gate: %1024==0)
+      for (int q = wid; q < nq; q += 4) {
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)(
+                base + q * 1024 + lane * 16),
+            (__attribute__((address_space(3))) void*)(dst + q * 1024),
+            16, 0, 0);
+      }
+    }
+    Floaded = target;
+  };
+
+  // ---- transposes -------------------------------------------------------
+  auto do_A = [&](int s) {  // s < 32: (ro = s>>2 co-octet, kg = s&3)
+    int ro = s >> 2, kg = s & 3;
+    __hip_bfloat16 t8[8][8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      // mask tail opix here (the glds fallback source is arbitrary
+      // valid memory, not zeros)
+      if (opix0 + kg * 8 + j < Mi) {
+        *reinterpret_cast<uint4*>(t8[j]) =
+            *reinterpret_cast<const uint4*>(
+                smem + RAW_A + (kg * 8 + j) * 128 + ro * 16);
+      } else {
+        *reinterpret_cast<uint4*>(t8[j]) = uint4{0, 0, 0, 0};
+      }
+    }
+#pragma unroll
+    for (int c = 0; c < 8; ++c) {
+      __hip_bfloat16 r[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) r[j] = t8[j][c];
+      *reinterpret_cast<uint4*>(
+          smem + IMG_A + lds_byte(ro * 8 + c, kg * 8)) =
+          *reinterpret_cast<const uint4*>(r);
+    }
+  };
+
+  auto do_B = [&](int s) {  // s in [0,288): ro = s>>2 tapci-octet, kg
+    int ro = s >> 2, kg = s & 3;
+    int c = ro >> 3;           // chunk 0..8
+    int dh = cdh[c], dw = cdw[c];
+    int ci0 = cci[c] + (ro & 7) * 8;
+    bool rok = cok[c];
+    // thread-local opix coords for its 8 opix (kg*8 + j)
+    int wo_ = wo0 + kg * 8;
+    int hr_ = 0;               // ho_rel (rows advanced past R0)
+    int ha_ = hoabs0;
+    while (wo_ >= Wo) {        // <= 1 iteration (Wo >= 28, kg*8 <= 24)
+      wo_ -= Wo;
+      ++hr_;
+      if (++ha_ == Ho) ha_ = 0;
+    }
+    int op_ = opix0 + kg * 8;
+    __hip_bfloat16 t8[8][8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      int hi = ha_ + dh - 1;
+      int wi = wo_ + dw - 1;
+      bool ok = rok && op_ + j < Mi && (unsigned)hi < (unsigned)Hi &&
+                (unsigned)wi < (unsigned)Wi;
+      int r5 = m5 + hr_ + dh - 1;
+      if (r5 < 0) r5 += WROWS;
+      if (r5 >= WROWS) r5 -= WROWS;
+      if (r5 >= WROWS) r5 -= WROWS;  // hr_+dh-1 can reach +5
+      const char* src = smem + WIN + r5 * rowpitch +
+                        (wi * Cin + ci0) * 2;
+      if (ok) {
+        *reinterpret_cast<uint4*>(t8[j]) =
+            *reinterpret_cast<const uint4*>(src);
+      } else {
+        *reinterpret_cast<uint4*>(t8[j]) = uint4{0, 0, 0, 0};
+      }
+      if (++wo_ == Wo) {
+        wo_ = 0;
+        ++hr_;
+        if (++ha_ == Ho) ha_ = 0;
+      }
+    }
+#pragma unroll
+    for (int cc = 0; cc < 8; ++cc) {
+      __hip_bfloat16 r[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) r[j] = t8[j][cc];
+      *reinterpret_cast<uint4*>(
+          smem + IMG_B + lds_byte(ro * 8 + cc, kg * 8)) =
+          *reinterpret_cast<const uint4*>(r);
+    }
+  };
+
+  auto advance_tile = [&]() {
+    opix0 += BK;
+    opA += BK;
+    wo0 += BK;
+    while (wo0 >= Wo) {
+      wo0 -= Wo;
+      ++R0;
+      if (++hoabs0 == Ho) hoabs0 = 0;
+      if (++m5 == WROWS) m5 = 0;
+    }
+  };
+
+  f32x4 acc[MREP][NREP] = {};
+  if (kt0 < kt1) {
+    issue_A();
+    // initial window: rows [R0-1 .. R1+1]
+    int R1 = R0 + (wo0 + BK - 1) / Wo;
+    load_rows(R1 + 1);
+    __syncthreads();  // drains glds (vmcnt0 implied)
+  }
+  for (int t = kt0; t < kt1; ++t) {
+    // transpose tile t out of rawA + ring window
+    {
+      int s = threadIdx.x;
+      if (s < 32) do_A(s);
+      else do_B(s - 32);
+      if (threadIdx.x < 64) do_B(256 - 32 + threadIdx.x);
+    }
+    __syncthreads();  // images ready; rawA free; old ring row reusable
+    if (t + 1 < kt1) {
+      advance_tile();
+      issue_A();
+      int R1 = R0 + (wo0 + BK - 1) / Wo;
+      load_rows(R1 + 1);  // 0-2 new rows, fly under the MFMAs
+    }
+    {
+      bf16x8 a_frag[MREP];
+      int kf = (lane >> 4) * 8;
+      int rowf = lane & 15;
+#pragma unroll
+      for (int mi = 0; mi < MREP; ++mi)
+        a_frag[mi] = *reinterpret_cast<const bf16x8*>(
+            smem + IMG_A + lds_byte(mi * 16 + rowf, kf));
+#pragma unroll
+      for (int ni = 0; ni < NREP; ++ni) {
+        bf16x8 b_frag = *reinterpret_cast<const bf16x8*>(
+            smem + IMG_B + lds_byte(wid * WN + ni * 16 + rowf, kf));
+#pragma unroll
+        for (int mi = 0; mi < MREP; ++mi)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag, acc[mi][ni], 0, 0, 0);
+      }
+    }
+    __syncthreads();  // images consumed + next glds drained
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MREP; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < NREP; ++ni) {
+      int col = tile_n + wid * WN + ni * 16 + (lane & 15);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = tile_m + mi * 16 + (lane >> 4) * 4 + j;
+        partial[(int64_t)row * Np + col] = acc[mi][ni][j];
+      }
+    }
+  }
+}
+
 // fp32 output: the split-K accumulation is fp32 and the consumer is the
 // fp32 master-weight gradient — rounding to bf16 here would be a
 // systematic numerics divergence vs the reference autocast path
@@ -1047,6 +1320,57 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
     force_v3 = (e && e[0] == '3') ? 1 : 0;
   }
   auto stream = at::hip::getCurrentHIPStream();
+
+  static int force_v4 = -1;
+  if (force_v4 < 0) {
+    const char* e = getenv("TURBOPRUNE_WRW");
+    force_v4 = (e && e[0] == '4') ? 1 : 0;
+  }
+  bool v5_ok = KH == 3 && KW == 3 && stride == 1 && pad == 1 &&
+               Hi == Wi && Ho == Wo && Hi == Ho && Wo >= 28 &&
+               (Wi * Cin * 2) % 1024 == 0 && Wi * Cin * 2 <= 7168;
+  if (!use_db && !use_v1 && !force_v2 && !force_v3 && !force_v4 &&
+      v5_ok) {
+    // v5: persistent-ring-window (contiguous glds, x read ~once)
+    constexpr int BM5 = conv_wrw5::BM, BN5 = conv_wrw5::BN,
+                  BK5 = conv_wrw5::BK;
+    int gm = (Cout + BM5 - 1) / BM5;
+    int gn = (K + BN5 - 1) / BN5;
+    int Mp5 = gm * BM5;
+    int Np5 = gn * BN5;
+    int tiles5 = gm * gn;
+    int total_kt = (int)((M + BK5 - 1) / BK5);
+    int splitk = 1;
+    while (tiles5 * splitk < 1024 && splitk * 2 <= total_kt &&
+           splitk < 512)
+      splitk *= 2;
+    auto partial = at::empty({splitk, (int64_t)Mp5, (int64_t)Np5},
+                             x.options().dtype(at::kFloat));
+    constexpr int kLds5 = 80 * 1024;
+    static bool attr5 = false;
+    if (!attr5) {
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(conv_wrw5_kernel),
+          hipFuncAttributeMaxDynamicSharedMemorySize, kLds5);
+      attr5 = true;
+    }
+    hipLaunchKernelGGL(conv_wrw5_kernel, dim3(tiles5, splitk), dim3(256),
+                       kLds5, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(gy.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       partial.data_ptr<float>(), Nb, Hi, Cin, Cout, Ho,
+                       Mp5, Np5, gn);
+    auto gw = at::empty({Cout, Cin, KH, KW},
+                        gy.options().dtype(at::kFloat).memory_format(
+                            at::MemoryFormat::ChannelsLast));
+    int64_t mn = (int64_t)Cout * K;
+    int rgrid = elementwise_grid(mn, kBlock, 4);
+    hipLaunchKernelGGL(wrw_reduce_kernel, dim3(rgrid), dim3(kBlock), 0,
+                       stream, partial.data_ptr<float>(),
+                       gw.data_ptr<float>(), mn,
+                       (int64_t)Mp5 * Np5, splitk, Np5, K, (int64_t)K);
+    return gw;
+  }
 
   if (!use_db && !use_v1 && !force_v2 && !force_v3 && KH == 3 &&
       KW == 3) {
