@@ -1023,6 +1023,9 @@ TP_DEVICE int lds_byte(int row, int k) {
 }
 }  // namespace conv_wrw5
 
+// MODE: 0 = normal; diagnostic ablations (results invalid):
+// 1 = no MFMA, 2 = no transpose, 3 = no glds (TURBOPRUNE_WRW5_MODE)
+template <int MODE = 0>
 __global__ __launch_bounds__(256) void conv_wrw5_kernel(
     const __hip_bfloat16* __restrict__ gy,  // (M, Cout) row-major
     const __hip_bfloat16* __restrict__ x,   // NHWC
@@ -1218,7 +1221,7 @@ __global__ __launch_bounds__(256) void conv_wrw5_kernel(
   }
   for (int t = kt0; t < kt1; ++t) {
     // transpose tile t out of rawA + ring window
-    {
+    if (MODE != 2) {
       int s = threadIdx.x;
       if (s < 32) do_A(s);
       else do_B(s - 32);
@@ -1227,11 +1230,13 @@ __global__ __launch_bounds__(256) void conv_wrw5_kernel(
     __syncthreads();  // images ready; rawA free; old ring row reusable
     if (t + 1 < kt1) {
       advance_tile();
-      issue_A();
-      int R1 = R0 + (wo0 + BK - 1) / Wo;
-      load_rows(R1 + 1);  // 0-2 new rows, fly under the MFMAs
+      if (MODE != 3) {
+        issue_A();
+        int R1 = R0 + (wo0 + BK - 1) / Wo;
+        load_rows(R1 + 1);  // 0-2 new rows, fly under the MFMAs
+      }
     }
-    {
+    if (MODE != 1) {
       bf16x8 a_frag[MREP];
       int kf = (lane >> 4) * 8;
       int rowf = lane & 15;
@@ -1347,14 +1352,22 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
     auto partial = at::empty({splitk, (int64_t)Mp5, (int64_t)Np5},
                              x.options().dtype(at::kFloat));
     constexpr int kLds5 = 80 * 1024;
-    static bool attr5 = false;
-    if (!attr5) {
-      (void)hipFuncSetAttribute(
-          reinterpret_cast<const void*>(conv_wrw5_kernel),
-          hipFuncAttributeMaxDynamicSharedMemorySize, kLds5);
-      attr5 = true;
+    static int mode5 = -1;
+    if (mode5 < 0) {
+      const char* e = getenv("TURBOPRUNE_WRW5_MODE");
+      mode5 = e ? atoi(e) : 0;
+      for (auto* f : {reinterpret_cast<const void*>(conv_wrw5_kernel<0>),
+                      reinterpret_cast<const void*>(conv_wrw5_kernel<1>),
+                      reinterpret_cast<const void*>(conv_wrw5_kernel<2>),
+                      reinterpret_cast<const void*>(conv_wrw5_kernel<3>)})
+        (void)hipFuncSetAttribute(
+            f, hipFuncAttributeMaxDynamicSharedMemorySize, kLds5);
     }
-    hipLaunchKernelGGL(conv_wrw5_kernel, dim3(tiles5, splitk), dim3(256),
+    auto* kfn = mode5 == 1 ? conv_wrw5_kernel<1>
+                : mode5 == 2 ? conv_wrw5_kernel<2>
+                : mode5 == 3 ? conv_wrw5_kernel<3>
+                             : conv_wrw5_kernel<0>;
+    hipLaunchKernelGGL(kfn, dim3(tiles5, splitk), dim3(256),
                        kLds5, stream,
                        reinterpret_cast<const __hip_bfloat16*>(gy.data_ptr()),
                        reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
