@@ -16,6 +16,8 @@
 #include <ATen/ATen.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <algorithm>
+
 #include "common.h"
 
 namespace turboprune {
@@ -1024,7 +1026,9 @@ TP_DEVICE int lds_byte(int row, int k) {
 }  // namespace conv_wrw5
 
 // MODE: 0 = normal; diagnostic ablations (results invalid):
-// 1 = no MFMA, 2 = no transpose, 3 = no glds (TURBOPRUNE_WRW5_MODE)
+// 1 = no MFMA, 2 = no transpose, 3 = no glds,
+// 4 = barriers+bookkeeping only, 5 = bookkeeping only
+// (TURBOPRUNE_WRW5_MODE)
 template <int MODE = 0>
 __global__ __launch_bounds__(256) void conv_wrw5_kernel(
     const __hip_bfloat16* __restrict__ gy,  // (M, Cout) row-major
@@ -1221,22 +1225,22 @@ __global__ __launch_bounds__(256) void conv_wrw5_kernel(
   }
   for (int t = kt0; t < kt1; ++t) {
     // transpose tile t out of rawA + ring window
-    if (MODE != 2) {
+    if (MODE != 2 && MODE < 4) {
       int s = threadIdx.x;
       if (s < 32) do_A(s);
       else do_B(s - 32);
       if (threadIdx.x < 64) do_B(256 - 32 + threadIdx.x);
     }
-    __syncthreads();  // images ready; rawA free; old ring row reusable
+    if (MODE != 5) __syncthreads();
     if (t + 1 < kt1) {
       advance_tile();
-      if (MODE != 3) {
+      if (MODE != 3 && MODE < 4) {
         issue_A();
         int R1 = R0 + (wo0 + BK - 1) / Wo;
         load_rows(R1 + 1);  // 0-2 new rows, fly under the MFMAs
       }
     }
-    if (MODE != 1) {
+    if (MODE != 1 && MODE < 4) {
       bf16x8 a_frag[MREP];
       int kf = (lane >> 4) * 8;
       int rowf = lane & 15;
@@ -1254,7 +1258,7 @@ __global__ __launch_bounds__(256) void conv_wrw5_kernel(
               a_frag[mi], b_frag, acc[mi][ni], 0, 0, 0);
       }
     }
-    __syncthreads();  // images consumed + next glds drained
+    if (MODE != 5) __syncthreads();  // images consumed + glds drained
   }
 
 #pragma unroll
@@ -1276,6 +1280,43 @@ __global__ __launch_bounds__(256) void conv_wrw5_kernel(
 // systematic numerics divergence vs the reference autocast path
 // (ADVICE r01). The output is only Cout x (KH*KW*Cin), so the extra
 // bytes are negligible next to the partial slabs.
+// Slab-parallel two-stage reduce: the single-stage kernel below sizes
+// its grid by mn (the OUTPUT, e.g. 36,864 elements for layer1) — at
+// splitk=512 that is 36 blocks streaming 75 MB with 512 strided reads
+// per element: ~450 us of pure latency, 52% of the whole v5 wrw call
+// (r2o mode-5 ablation: the "empty" kernel + reduce cost 491 us).
+// Stage 1 spreads GROUPS of slabs across gridDim.y (full chip);
+// stage 2 folds the GROUPS (deterministic, no atomics).
+__global__ void wrw_reduce_stage1_kernel(
+    const float* __restrict__ partial, float* __restrict__ mid,
+    int64_t mn_padded, int64_t slab_stride, int slabs, int groups) {
+  int g = blockIdx.y;
+  int s0 = (int)((int64_t)g * slabs / groups);
+  int s1 = (int)((int64_t)(g + 1) * slabs / groups);
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < mn_padded;
+       i += stride) {
+    float v = 0.f;
+    for (int s = s0; s < s1; ++s) v += partial[s * slab_stride + i];
+    mid[(int64_t)g * mn_padded + i] = v;
+  }
+}
+
+__global__ void wrw_reduce_stage2_kernel(
+    const float* __restrict__ mid, float* __restrict__ out, int64_t mn,
+    int64_t mn_padded, int groups, int Np, int64_t out_cols) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < mn;
+       i += stride) {
+    int64_t row = i / out_cols;
+    int64_t col = i % out_cols;
+    int64_t src = row * Np + col;
+    float v = 0.f;
+    for (int g = 0; g < groups; ++g) v += mid[(int64_t)g * mn_padded + src];
+    out[i] = v;
+  }
+}
+
 __global__ void wrw_reduce_kernel(const float* __restrict__ partial,
                                   float* __restrict__ out,
                                   int64_t mn, int64_t slab_stride,
@@ -1290,6 +1331,35 @@ __global__ void wrw_reduce_kernel(const float* __restrict__ partial,
     for (int s = 0; s < slabs; ++s)
       v += partial[(int64_t)s * slab_stride + row * Np + col];
     out[i] = v;
+  }
+}
+
+// launches the right reduce for (slabs, mn): slab-parallel two-stage
+// when the single-stage grid would underfill the chip
+static void launch_wrw_reduce(const at::Tensor& partial, at::Tensor& gw,
+                              int64_t Mp, int64_t Np, int slabs, int K,
+                              int Cout, hipStream_t stream) {
+  int64_t mn = (int64_t)Cout * K;
+  int64_t mn_padded = Mp * Np;
+  int rgrid = elementwise_grid(mn, kBlock, 4);
+  if (slabs >= 32 && rgrid < 512) {
+    int groups = std::min(32, slabs);
+    auto mid = at::empty({groups, mn_padded},
+                         partial.options().dtype(at::kFloat));
+    int g1 = elementwise_grid(mn_padded, kBlock, 4);
+    hipLaunchKernelGGL(wrw_reduce_stage1_kernel, dim3(g1, groups),
+                       dim3(kBlock), 0, stream,
+                       partial.data_ptr<float>(), mid.data_ptr<float>(),
+                       mn_padded, mn_padded, slabs, groups);
+    hipLaunchKernelGGL(wrw_reduce_stage2_kernel, dim3(rgrid),
+                       dim3(kBlock), 0, stream, mid.data_ptr<float>(),
+                       gw.data_ptr<float>(), mn, mn_padded, groups,
+                       (int)Np, (int64_t)K);
+  } else {
+    hipLaunchKernelGGL(wrw_reduce_kernel, dim3(rgrid), dim3(kBlock), 0,
+                       stream, partial.data_ptr<float>(),
+                       gw.data_ptr<float>(), mn, mn_padded, slabs,
+                       (int)Np, K, (int64_t)K);
   }
 }
 
@@ -1359,13 +1429,17 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
       for (auto* f : {reinterpret_cast<const void*>(conv_wrw5_kernel<0>),
                       reinterpret_cast<const void*>(conv_wrw5_kernel<1>),
                       reinterpret_cast<const void*>(conv_wrw5_kernel<2>),
-                      reinterpret_cast<const void*>(conv_wrw5_kernel<3>)})
+                      reinterpret_cast<const void*>(conv_wrw5_kernel<3>),
+                      reinterpret_cast<const void*>(conv_wrw5_kernel<4>),
+                      reinterpret_cast<const void*>(conv_wrw5_kernel<5>)})
         (void)hipFuncSetAttribute(
             f, hipFuncAttributeMaxDynamicSharedMemorySize, kLds5);
     }
     auto* kfn = mode5 == 1 ? conv_wrw5_kernel<1>
                 : mode5 == 2 ? conv_wrw5_kernel<2>
                 : mode5 == 3 ? conv_wrw5_kernel<3>
+                : mode5 == 4 ? conv_wrw5_kernel<4>
+                : mode5 == 5 ? conv_wrw5_kernel<5>
                              : conv_wrw5_kernel<0>;
     hipLaunchKernelGGL(kfn, dim3(tiles5, splitk), dim3(256),
                        kLds5, stream,
@@ -1376,12 +1450,7 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
     auto gw = at::empty({Cout, Cin, KH, KW},
                         gy.options().dtype(at::kFloat).memory_format(
                             at::MemoryFormat::ChannelsLast));
-    int64_t mn = (int64_t)Cout * K;
-    int rgrid = elementwise_grid(mn, kBlock, 4);
-    hipLaunchKernelGGL(wrw_reduce_kernel, dim3(rgrid), dim3(kBlock), 0,
-                       stream, partial.data_ptr<float>(),
-                       gw.data_ptr<float>(), mn,
-                       (int64_t)Mp5 * Np5, splitk, Np5, K, (int64_t)K);
+    launch_wrw_reduce(partial, gw, Mp5, Np5, splitk, K, Cout, stream);
     return gw;
   }
 
@@ -1425,12 +1494,7 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
     auto gw = at::empty({Cout, Cin, KH, KW},
                         gy.options().dtype(at::kFloat).memory_format(
                             at::MemoryFormat::ChannelsLast));
-    int64_t mn = (int64_t)Cout * K;
-    int rgrid = elementwise_grid(mn, kBlock, 4);
-    hipLaunchKernelGGL(wrw_reduce_kernel, dim3(rgrid), dim3(kBlock), 0,
-                       stream, partial.data_ptr<float>(),
-                       gw.data_ptr<float>(), mn,
-                       (int64_t)Mp4 * Np4, splitk, Np4, K, (int64_t)K);
+    launch_wrw_reduce(partial, gw, Mp4, Np4, splitk, K, Cout, stream);
     return gw;
   }
 
@@ -1459,12 +1523,7 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
     auto gw = at::empty({Cout, Cin, KH, KW},
                         gy.options().dtype(at::kFloat).memory_format(
                             at::MemoryFormat::ChannelsLast));
-    int64_t mn = (int64_t)Cout * K;
-    int rgrid = elementwise_grid(mn, kBlock, 4);
-    hipLaunchKernelGGL(wrw_reduce_kernel, dim3(rgrid), dim3(kBlock), 0,
-                       stream, partial.data_ptr<float>(),
-                       gw.data_ptr<float>(), mn,
-                       (int64_t)Mp3 * Np3, splitk, Np3, K, (int64_t)K);
+    launch_wrw_reduce(partial, gw, Mp3, Np3, splitk, K, Cout, stream);
     return gw;
   }
 
@@ -1486,12 +1545,7 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
     auto gw = at::empty({Cout, Cin, KH, KW},
                         gy.options().dtype(at::kFloat).memory_format(
                             at::MemoryFormat::ChannelsLast));
-    int64_t mn = (int64_t)Cout * K;
-    int rgrid = elementwise_grid(mn, kBlock, 4);
-    hipLaunchKernelGGL(wrw_reduce_kernel, dim3(rgrid), dim3(kBlock), 0,
-                       stream, partial.data_ptr<float>(),
-                       gw.data_ptr<float>(), mn,
-                       (int64_t)Mp * Np, splitk, Np, K, (int64_t)K);
+    launch_wrw_reduce(partial, gw, Mp, Np, splitk, K, Cout, stream);
     return gw;
   }
 
