@@ -1280,6 +1280,223 @@ __global__ __launch_bounds__(256) void conv_wrw5_kernel(
 // systematic numerics divergence vs the reference autocast path
 // (ADVICE r01). The output is only Cout x (KH*KW*Cin), so the extra
 // bytes are negligible next to the partial slabs.
+// ---- v6: glds-blocked image + ds_read_b64_tr_b16 fragments -------------
+//
+// v4/v5's measured wall is the LDS pipe: the raw->img transpose pass
+// moves ~140 KiB/tile through the 128 B/cycle port. CDNA4's
+// ds_read_b64_tr_b16 deletes that pass: each 16-lane quarter-group
+// reads one contiguous [4 k][16 col] bf16 block and receives its
+// TRANSPOSE (lane = column, 4 regs = rows; semantics probed on device,
+// scripts/probe_tr16.hip). Because the staging is a per-lane GATHER,
+// the glds deposit builds the blocked image directly (permuting which
+// (opix, ci-octet) each lane loads is free), so the kernel becomes:
+//
+//   glds(t+1 -> other raw buffer)  |  MFMA(t) with tr_b16 fragment
+//   reads straight from raw        |  one barrier per tile
+//
+// LDS: 2 x (A 4K + B 36K) = 80 KiB -> 2 blocks/CU. No transpose
+// stage, no img buffers, per-lane zero-page bounds as v4.
+//
+// Image layout (B): block(kq, cs) at ((chunk*8 + kq)*4 + cs)*128 B,
+// kq = opix quad (8 per 32-opix tile), cs = 16-tapci subtile within
+// the 64-tapci chunk; one glds instr (chunk, og) deposits its 8
+// blocks contiguously at (chunk*32 + og*8)*128 — identical dst
+// arithmetic to v4, different internal granule order. A analogous.
+namespace conv_wrw6 {
+constexpr int BM = 64, BN = 576, BK = 32;
+constexpr int WN = 144;
+constexpr int MREP = 4, NREP = 9;
+constexpr int RAW_A0 = 0;            //  4 KiB
+constexpr int RAW_B0 = 4 * 1024;     // 36 KiB
+constexpr int BUF_STRIDE = 40 * 1024;
+}  // namespace conv_wrw6
+
+// two tr_b16 reads (k-quads h2=0,1 at +512 B) -> one MFMA operand
+TP_DEVICE bf16x8 tr_frag16(unsigned a0) {
+  unsigned long long r0, r1;
+  asm volatile(
+      "ds_read_b64_tr_b16 %0, %2\n\t"
+      "ds_read_b64_tr_b16 %1, %2 offset:512\n\t"
+      "s_waitcnt lgkmcnt(0)"
+      : "=v"(r0), "=v"(r1)
+      : "v"(a0));
+  union {
+    unsigned long long u[2];
+    bf16x8 v;
+  } c;
+  c.u[0] = r0;
+  c.u[1] = r1;
+  return c.v;
+}
+
+__global__ __launch_bounds__(256) void conv_wrw6_kernel(
+    const __hip_bfloat16* __restrict__ gy,  // (M, Cout) row-major
+    const __hip_bfloat16* __restrict__ x,   // NHWC
+    const __hip_bfloat16* __restrict__ zero_page,  // >=128 B zeros
+    float* __restrict__ partial,            // (slabs, Mp, Np)
+    int Nb, int Hi, int Wi, int Cin, int Cout, int Ho, int Wo, int KH,
+    int KW, int stride, int pad, int Mp, int Np, int grid_n) {
+  using namespace conv_wrw6;
+  extern __shared__ char smem[];
+
+  int wg = blockIdx.x;
+  int tile_m = (wg / grid_n) * BM;  // co tile
+  int tile_n = (wg % grid_n) * BN;  // tapci tile
+  int64_t M64 = (int64_t)Nb * Ho * Wo;
+  int Mi = (int)M64;
+
+  int lane = threadIdx.x & (kWave - 1);
+  int wid = threadIdx.x / kWave;
+
+  int total_kt = (int)((M64 + BK - 1) / BK);
+  int per = (total_kt + gridDim.y - 1) / gridDim.y;
+  int kt0 = blockIdx.y * per;
+  int kt1 = min(kt0 + per, total_kt);
+  partial += (int64_t)blockIdx.y * Mp * Np;
+
+  // per-lane deposit geometry: granule l -> block b = l>>3 of the
+  // instr's 1 KiB (b = quad-half q * 4 + col-subtile cs), row-in-block
+  // r2 = (l&7)>>1, column-octet-half h = l&1. Lane's opix offset
+  // within the instr's 8-opix group: q*4 + r2.
+  int q_ = (lane >> 5) & 1;
+  int cs_ = (lane >> 3) & 3;
+  int r2_ = (lane & 7) >> 1;
+  int h_ = lane & 1;
+  int opoff = q_ * 4 + r2_;
+
+  int Ktot = KH * KW * Cin;
+  int cdh[9], cdw[9], cci[9];
+  bool cok[9];
+#pragma unroll
+  for (int c = 0; c < 9; ++c) {
+    int tapci = tile_n + c * 64;
+    cok[c] = tapci < Ktot;
+    int tap = cok[c] ? tapci / Cin : 0;
+    cci[c] = cok[c] ? tapci % Cin : 0;
+    cdh[c] = tap / KW;
+    cdw[c] = tap % KW;
+  }
+
+  // per-lane coordinate sets for opix = kt*BK + og*8 + opoff, og=0..3
+  int wo4[4], ho4[4], n4[4], op4[4];
+#pragma unroll
+  for (int og = 0; og < 4; ++og) {
+    int opix = kt0 * BK + og * 8 + opoff;
+    op4[og] = opix;
+    int o = opix < Mi ? opix : 0;
+    wo4[og] = o % Wo;
+    int r2v = o / Wo;
+    ho4[og] = r2v % Ho;
+    n4[og] = r2v / Ho;
+  }
+  auto advance = [&]() {
+#pragma unroll
+    for (int og = 0; og < 4; ++og) {
+      op4[og] += BK;
+      wo4[og] += BK;
+      while (wo4[og] >= Wo) {
+        wo4[og] -= Wo;
+        if (++ho4[og] == Ho) {
+          ho4[og] = 0;
+          ++n4[og];
+        }
+      }
+    }
+  };
+
+  auto issue_glds = [&](int buf) {
+    char* rawA = smem + buf * BUF_STRIDE + RAW_A0;
+    char* rawB = smem + buf * BUF_STRIDE + RAW_B0;
+    // A: wave wid's 8-opix group; granule = (co-octet cs_*2+h_,
+    // opix base + opoff)
+    {
+      int og = wid;
+      int co8 = cs_ * 2 + h_;
+      const char* src;
+      if (op4[og] < Mi) {
+        src = reinterpret_cast<const char*>(
+            gy + (int64_t)op4[og] * Cout + tile_m + co8 * 8);
+      } else {
+        src = reinterpret_cast<const char*>(zero_page) + co8 * 16;
+      }
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(rawA + og * 1024),
+          16, 0, 0);
+    }
+#pragma unroll
+    for (int i = 0; i < 9; ++i) {
+      int c = i;
+      int og = (i + wid) & 3;
+      int hi = ho4[og] * stride - pad + cdh[c];
+      int wi = wo4[og] * stride - pad + cdw[c];
+      int ci8 = cs_ * 2 + h_;  // ci octet within the 64-ci chunk
+      bool ok = cok[c] && op4[og] < Mi && hi >= 0 && hi < Hi &&
+                wi >= 0 && wi < Wi;
+      const char* src =
+          ok ? reinterpret_cast<const char*>(
+                   x + ((int64_t)(n4[og] * Hi + hi) * Wi + wi) * Cin +
+                   cci[c] + ci8 * 8)
+             : reinterpret_cast<const char*>(zero_page) + ci8 * 16;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)(
+              rawB + (c * 32 + og * 8) * 128),
+          16, 0, 0);
+    }
+  };
+
+  f32x4 acc[MREP][NREP] = {};
+  int cur = 0;
+  if (kt0 < kt1) {
+    issue_glds(0);
+    __syncthreads();  // carries vmcnt(0): deposits landed
+  }
+  for (int t = kt0; t < kt1; ++t) {
+    if (t + 1 < kt1) {
+      advance();
+      issue_glds(cur ^ 1);  // next tile into the other buffer
+    }
+    {
+      char* rawA = smem + cur * BUF_STRIDE + RAW_A0;
+      char* rawB = smem + cur * BUF_STRIDE + RAW_B0;
+      int g = lane >> 4;
+      int l15 = lane & 15;
+      bf16x8 a_frag[MREP];
+#pragma unroll
+      for (int mi = 0; mi < MREP; ++mi)
+        a_frag[mi] = tr_frag16((unsigned)(size_t)(
+            rawA + ((2 * g) * 4 + mi) * 128 + l15 * 8));
+#pragma unroll
+      for (int ni = 0; ni < NREP; ++ni) {
+        int cs_abs = wid * 9 + ni;          // 16-tapci subtile index
+        int c = cs_abs >> 2, cs = cs_abs & 3;
+        bf16x8 b_frag = tr_frag16((unsigned)(size_t)(
+            rawB + ((c * 8 + 2 * g) * 4 + cs) * 128 + l15 * 8));
+#pragma unroll
+        for (int mi = 0; mi < MREP; ++mi)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag, acc[mi][ni], 0, 0, 0);
+      }
+    }
+    __syncthreads();  // raw consumed + next deposits drained
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MREP; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < NREP; ++ni) {
+      int col = tile_n + wid * WN + ni * 16 + (lane & 15);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = tile_m + mi * 16 + (lane >> 4) * 4 + j;
+        partial[(int64_t)row * Np + col] = acc[mi][ni][j];
+      }
+    }
+  }
+}
+
 // Slab-parallel two-stage reduce: the single-stage kernel below sizes
 // its grid by mn (the OUTPUT, e.g. 36,864 elements for layer1) — at
 // splitk=512 that is 36 blocks streaming 75 MB with 512 strided reads
@@ -1401,11 +1618,63 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
     const char* e = getenv("TURBOPRUNE_WRW");
     force_v4 = (e && e[0] == '4') ? 1 : 0;
   }
+  static int force_v5 = -1, force_v6 = -1;
+  if (force_v5 < 0) {
+    const char* e = getenv("TURBOPRUNE_WRW");
+    force_v5 = (e && e[0] == '5') ? 1 : 0;
+    force_v6 = (e && e[0] == '6') ? 1 : 0;
+  }
+  if ((force_v6 ||
+       (!use_db && !use_v1 && !force_v2 && !force_v3 && !force_v4 &&
+        !force_v5)) &&
+      KH == 3 && KW == 3) {
+    // v6: tr_b16 fragments straight from the glds-blocked raw image
+    constexpr int BM6 = conv_wrw6::BM, BN6 = conv_wrw6::BN,
+                  BK6 = conv_wrw6::BK;
+    int gm = (Cout + BM6 - 1) / BM6;
+    int gn = (K + BN6 - 1) / BN6;
+    int Mp6 = gm * BM6;
+    int Np6 = gn * BN6;
+    int tiles6 = gm * gn;
+    int total_kt = (int)((M + BK6 - 1) / BK6);
+    int splitk = 1;
+    while (tiles6 * splitk < 1024 && splitk * 2 <= total_kt &&
+           splitk < 512)
+      splitk *= 2;
+    auto partial = at::empty({splitk, (int64_t)Mp6, (int64_t)Np6},
+                             x.options().dtype(at::kFloat));
+    static at::Tensor zp6;
+    if (!zp6.defined() || zp6.device() != x.device())
+      zp6 = at::zeros({64}, x.options());
+    constexpr int kLds6 = 80 * 1024;
+    static bool attr6 = false;
+    if (!attr6) {
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(conv_wrw6_kernel),
+          hipFuncAttributeMaxDynamicSharedMemorySize, kLds6);
+      attr6 = true;
+    }
+    hipLaunchKernelGGL(conv_wrw6_kernel, dim3(tiles6, splitk), dim3(256),
+                       kLds6, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(gy.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(
+                           zp6.data_ptr()),
+                       partial.data_ptr<float>(), Nb, Hi, Wi, Cin, Cout,
+                       Ho, Wo, (int)KH, (int)KW, (int)stride, (int)pad,
+                       Mp6, Np6, gn);
+    auto gw = at::empty({Cout, Cin, KH, KW},
+                        gy.options().dtype(at::kFloat).memory_format(
+                            at::MemoryFormat::ChannelsLast));
+    launch_wrw_reduce(partial, gw, Mp6, Np6, splitk, K, Cout, stream);
+    return gw;
+  }
+
   bool v5_ok = KH == 3 && KW == 3 && stride == 1 && pad == 1 &&
                Hi == Wi && Ho == Wo && Hi == Ho && Wo >= 28 &&
                (Wi * Cin * 2) % 1024 == 0 && Wi * Cin * 2 <= 7168;
   if (!use_db && !use_v1 && !force_v2 && !force_v3 && !force_v4 &&
-      v5_ok) {
+      force_v5 && v5_ok) {
     // v5: persistent-ring-window (contiguous glds, x read ~once)
     constexpr int BM5 = conv_wrw5::BM, BN5 = conv_wrw5::BN,
                   BK5 = conv_wrw5::BK;
