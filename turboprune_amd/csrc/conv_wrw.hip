@@ -1311,22 +1311,39 @@ constexpr int RAW_B0 = 4 * 1024;     // 36 KiB
 constexpr int BUF_STRIDE = 40 * 1024;
 }  // namespace conv_wrw6
 
-// two tr_b16 reads (k-quads h2=0,1 at +512 B) -> one MFMA operand
-TP_DEVICE bf16x8 tr_frag16(unsigned a0) {
+// two tr_b16 reads (k-quads h2=0,1 at +512 B) -> one MFMA operand.
+// Split issue/wait so reads pipeline under MFMAs: tr_issue starts the
+// pair, tr_wait drains lgkm and carries a register dependency so the
+// compiler cannot move uses above it.
+struct TrPair {
   unsigned long long r0, r1;
+};
+TP_DEVICE void tr_issue(unsigned a0, TrPair& p) {
   asm volatile(
       "ds_read_b64_tr_b16 %0, %2\n\t"
-      "ds_read_b64_tr_b16 %1, %2 offset:512\n\t"
-      "s_waitcnt lgkmcnt(0)"
-      : "=v"(r0), "=v"(r1)
+      "ds_read_b64_tr_b16 %1, %2 offset:512"
+      : "=v"(p.r0), "=v"(p.r1)
       : "v"(a0));
+}
+// N = DS ops allowed to stay outstanding (DS completes in order, so
+// lgkmcnt(2) proves an older pair landed while a newer pair flies)
+template <int N>
+TP_DEVICE bf16x8 tr_wait(TrPair& p) {
+  asm volatile("s_waitcnt lgkmcnt(%c2)"
+               : "+v"(p.r0), "+v"(p.r1)
+               : "i"(N));
   union {
     unsigned long long u[2];
     bf16x8 v;
   } c;
-  c.u[0] = r0;
-  c.u[1] = r1;
+  c.u[0] = p.r0;
+  c.u[1] = p.r1;
   return c.v;
+}
+TP_DEVICE bf16x8 tr_frag16(unsigned a0) {
+  TrPair p;
+  tr_issue(a0, p);
+  return tr_wait<0>(p);
 }
 
 __global__ __launch_bounds__(256) void conv_wrw6_kernel(
@@ -1462,21 +1479,35 @@ __global__ __launch_bounds__(256) void conv_wrw6_kernel(
       char* rawB = smem + cur * BUF_STRIDE + RAW_B0;
       int g = lane >> 4;
       int l15 = lane & 15;
+      auto baddr = [&](int ni) {
+        int cs_abs = wid * 9 + ni;          // 16-tapci subtile index
+        int c = cs_abs >> 2, cs = cs_abs & 3;
+        return (unsigned)(size_t)(
+            rawB + ((c * 8 + 2 * g) * 4 + cs) * 128 + l15 * 8);
+      };
+      // pipeline: all 4 A pairs + B0 issued up front (one drain), each
+      // B(ni+1) issued before ni's MFMAs so its latency hides
+      TrPair pa[MREP], pb0, pb1;
+#pragma unroll
+      for (int mi = 0; mi < MREP; ++mi)
+        tr_issue((unsigned)(size_t)(
+                     rawA + ((2 * g) * 4 + mi) * 128 + l15 * 8),
+                 pa[mi]);
+      tr_issue(baddr(0), pb0);
       bf16x8 a_frag[MREP];
 #pragma unroll
       for (int mi = 0; mi < MREP; ++mi)
-        a_frag[mi] = tr_frag16((unsigned)(size_t)(
-            rawA + ((2 * g) * 4 + mi) * 128 + l15 * 8));
+        a_frag[mi] = tr_wait<2>(pa[mi]);  // pb0 may stay in flight
 #pragma unroll
       for (int ni = 0; ni < NREP; ++ni) {
-        int cs_abs = wid * 9 + ni;          // 16-tapci subtile index
-        int c = cs_abs >> 2, cs = cs_abs & 3;
-        bf16x8 b_frag = tr_frag16((unsigned)(size_t)(
-            rawB + ((c * 8 + 2 * g) * 4 + cs) * 128 + l15 * 8));
+        if (ni + 1 < NREP) tr_issue(baddr(ni + 1), pb1);
+        bf16x8 b_frag =
+            ni + 1 < NREP ? tr_wait<2>(pb0) : tr_wait<0>(pb0);
 #pragma unroll
         for (int mi = 0; mi < MREP; ++mi)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a_frag[mi], b_frag, acc[mi][ni], 0, 0, 0);
+        pb0 = pb1;
       }
     }
     __syncthreads();  // raw consumed + next deposits drained
