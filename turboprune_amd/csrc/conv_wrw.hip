@@ -1382,20 +1382,32 @@ __global__ __launch_bounds__(256) void conv_wrw6_kernel(
   int opoff = q_ * 4 + r2_;
 
   int Ktot = KH * KW * Cin;
-  int cdh[9], cdw[9], cci[9];
+  // per-chunk constants: tap byte offset into x relative to the
+  // (n, ho*stride-pad, wo*stride-pad) corner, ci-chunk byte offset,
+  // and the (dh, dw) needed for bounds
+  int cdh[9], cdw[9];
+  int coff[9];  // ((dh*Wi + dw)*Cin + ci0) * 2
   bool cok[9];
 #pragma unroll
   for (int c = 0; c < 9; ++c) {
     int tapci = tile_n + c * 64;
     cok[c] = tapci < Ktot;
     int tap = cok[c] ? tapci / Cin : 0;
-    cci[c] = cok[c] ? tapci % Cin : 0;
+    int ci0 = cok[c] ? tapci % Cin : 0;
     cdh[c] = tap / KW;
     cdw[c] = tap % KW;
+    coff[c] = ((cdh[c] * Wi + cdw[c]) * Cin + ci0) * 2;
   }
 
-  // per-lane coordinate sets for opix = kt*BK + og*8 + opoff, og=0..3
-  int wo4[4], ho4[4], n4[4], op4[4];
+  // per-lane coordinate sets for opix = kt*BK + og*8 + opoff, og=0..3.
+  // Besides (wo, ho, n) each og carries the 32-bit BYTE offset of its
+  // corner x element ((n*Hi + ho*s - p)*Wi + wo*s - p)*Cin*2 — updated
+  // incrementally so the per-instr address math is one add
+  // (PMC r2q: the per-instr 64-bit address formation made
+  // SQ_INSTS_VALU ~= MFMA busy cycles).
+  int wo4[4], ho4[4], n4[4], op4[4], xo4[4];
+  const int stepw = stride * Cin * 2;                    // wo+1
+  const int rowfix_num = (stride * Wi - 0) * Cin * 2;    // see advance
 #pragma unroll
   for (int og = 0; og < 4; ++og) {
     int opix = kt0 * BK + og * 8 + opoff;
@@ -1405,17 +1417,25 @@ __global__ __launch_bounds__(256) void conv_wrw6_kernel(
     int r2v = o / Wo;
     ho4[og] = r2v % Ho;
     n4[og] = r2v / Ho;
+    xo4[og] = (((n4[og] * Hi + ho4[og] * stride - pad) * Wi) +
+               wo4[og] * stride - pad) * Cin * 2;
   }
   auto advance = [&]() {
 #pragma unroll
     for (int og = 0; og < 4; ++og) {
       op4[og] += BK;
       wo4[og] += BK;
+      xo4[og] += BK * stepw;
       while (wo4[og] >= Wo) {
         wo4[og] -= Wo;
-        if (++ho4[og] == Ho) {
+        ho4[og] += 1;
+        // corner moved: -Wo columns, +1 output row (stride rows of x)
+        xo4[og] += rowfix_num - Wo * stepw;
+        if (ho4[og] == Ho) {
           ho4[og] = 0;
           ++n4[og];
+          // -Ho output rows, +1 image
+          xo4[og] += (Hi - Ho * stride) * Wi * Cin * 2;
         }
       }
     }
@@ -1441,20 +1461,19 @@ __global__ __launch_bounds__(256) void conv_wrw6_kernel(
           (__attribute__((address_space(3))) void*)(rawA + og * 1024),
           16, 0, 0);
     }
+    const char* xb = reinterpret_cast<const char*>(x) +
+                     (int64_t)(cs_ * 2 + h_) * 16;
+    const char* zb = reinterpret_cast<const char*>(zero_page) +
+                     (cs_ * 2 + h_) * 16;
 #pragma unroll
     for (int i = 0; i < 9; ++i) {
       int c = i;
       int og = (i + wid) & 3;
       int hi = ho4[og] * stride - pad + cdh[c];
       int wi = wo4[og] * stride - pad + cdw[c];
-      int ci8 = cs_ * 2 + h_;  // ci octet within the 64-ci chunk
-      bool ok = cok[c] && op4[og] < Mi && hi >= 0 && hi < Hi &&
-                wi >= 0 && wi < Wi;
-      const char* src =
-          ok ? reinterpret_cast<const char*>(
-                   x + ((int64_t)(n4[og] * Hi + hi) * Wi + wi) * Cin +
-                   cci[c] + ci8 * 8)
-             : reinterpret_cast<const char*>(zero_page) + ci8 * 16;
+      bool ok = cok[c] && op4[og] < Mi && (unsigned)hi < (unsigned)Hi &&
+                (unsigned)wi < (unsigned)Wi;
+      const char* src = ok ? xb + xo4[og] + coff[c] : zb;
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) void*)src,
           (__attribute__((address_space(3))) void*)(
