@@ -1302,14 +1302,6 @@ __global__ __launch_bounds__(256) void conv_wrw5_kernel(
 // the 64-tapci chunk; one glds instr (chunk, og) deposits its 8
 // blocks contiguously at (chunk*32 + og*8)*128 — identical dst
 // arithmetic to v4, different internal granule order. A analogous.
-namespace conv_wrw6 {
-constexpr int BM = 64, BN = 576, BK = 32;
-constexpr int WN = 144;
-constexpr int MREP = 4, NREP = 9;
-constexpr int RAW_A0 = 0;            //  4 KiB
-constexpr int RAW_B0 = 4 * 1024;     // 36 KiB
-constexpr int BUF_STRIDE = 40 * 1024;
-}  // namespace conv_wrw6
 
 // two tr_b16 reads (k-quads h2=0,1 at +512 B) -> one MFMA operand.
 // Split issue/wait so reads pipeline under MFMAs: tr_issue starts the
@@ -1345,6 +1337,301 @@ TP_DEVICE bf16x8 tr_frag16(unsigned a0) {
   tr_issue(a0, p);
   return tr_wait<0>(p);
 }
+
+// single-read variants (one k-quad = 4 k per lane)
+TP_DEVICE void tr_issue1(unsigned a0, unsigned long long& r) {
+  asm volatile("ds_read_b64_tr_b16 %0, %1" : "=v"(r) : "v"(a0));
+}
+template <int N>
+TP_DEVICE unsigned long long tr_wait1(unsigned long long& r) {
+  asm volatile("s_waitcnt lgkmcnt(%c1)" : "+v"(r) : "i"(N));
+  return r;
+}
+
+
+// ---- v7: ring window + tr_b16 fragments straight from the window -------
+//
+// The tr_b16 "run" (4 contiguous bf16 at a per-lane address) for a
+// B-fragment element is 4 consecutive ci of one (opix, tap) — which
+// exists VERBATIM inside the v5 ring window (x rows resident in LDS).
+// So for the large-Wo stride-1 3x3 shapes the kernel needs neither a
+// blocked image nor a transpose: contiguous ~7 KiB row loads per tile
+// (x read ~once) and the MFMA B-operands gather from the window with
+// per-lane run addresses. A stays on the v6 gather+blocked-tr path
+// (tiny). One barrier per tile; out-of-range taps select a zeroed LDS
+// block. WROWS=7 makes the t+1 row prefetch provably non-colliding
+// with tile t's reads (max flat-row advance is 2 at Wo>=28).
+namespace conv_wrw7 {
+constexpr int BM = 64, BN = 576, BK = 32;
+constexpr int WN = 144;
+constexpr int MREP = 4, NREP = 9;
+constexpr int WROWS = 7;
+constexpr int MAXROW = 7168;                  // bytes (gate)
+constexpr int WIN = 0;                        // 7 x rowpitch
+constexpr int RAW_A0 = WROWS * MAXROW;        // 2 x 4 KiB
+constexpr int ZBLK = RAW_A0 + 2 * 4096;       // 128 B zeros
+constexpr int LDS_BYTES = ZBLK + 128;
+}  // namespace conv_wrw7
+
+__global__ __launch_bounds__(256) void conv_wrw7_kernel(
+    const __hip_bfloat16* __restrict__ gy,  // (M, Cout) row-major
+    const __hip_bfloat16* __restrict__ x,   // NHWC
+    const __hip_bfloat16* __restrict__ zero_page,
+    float* __restrict__ partial,            // (slabs, Mp, Np)
+    int Nb, int HW, int Cin, int Cout, int Mp, int Np, int grid_n) {
+  using namespace conv_wrw7;
+  extern __shared__ char smem[];
+  const int Wo = HW, Ho = HW, Hi = HW, Wi = HW;  // k3 s1 p1 square
+  const int rowpitch = Wi * Cin * 2;
+
+  int wg = blockIdx.x;
+  int tile_m = (wg / grid_n) * BM;
+  int tile_n = (wg % grid_n) * BN;
+  int64_t M64 = (int64_t)Nb * Ho * Wo;
+  int Mi = (int)M64;
+  int Frows = Nb * Hi;
+
+  int lane = threadIdx.x & (kWave - 1);
+  int wid = threadIdx.x / kWave;
+  int g = lane >> 4;
+  int l15 = lane & 15;
+  int jl = l15 >> 2;   // fragment row within quad
+  int ml = l15 & 3;    // run (4-ci quad) within subtile
+
+  int total_kt = (int)((M64 + BK - 1) / BK);
+  int per = (total_kt + gridDim.y - 1) / gridDim.y;
+  int kt0 = blockIdx.y * per;
+  int kt1 = min(kt0 + per, total_kt);
+  partial += (int64_t)blockIdx.y * Mp * Np;
+
+  // zero the LDS zero-block (any wave; barrier below orders it)
+  if (threadIdx.x < 16)
+    reinterpret_cast<long long*>(smem + ZBLK)[threadIdx.x] = 0;
+
+  // per-wave per-ni B constants: 16-tapci subtile -> (dh, dw, ci0);
+  // noff folds (dw-1)*Cin + ci0 + 4*ml into one byte constant
+  int tdh[NREP], tdw[NREP];
+  int noff[NREP];
+  int Ktot = 9 * Cin;
+  bool nok[NREP];
+#pragma unroll
+  for (int ni = 0; ni < NREP; ++ni) {
+    int tapci0 = (wid * 9 + ni) * 16;
+    nok[ni] = tile_n + tapci0 < Ktot;
+    int tap = nok[ni] ? (tile_n + tapci0) / Cin : 0;
+    int ci0 = nok[ni] ? (tile_n + tapci0) % Cin : 0;
+    tdh[ni] = tap / 3;
+    tdw[ni] = tap % 3;
+    noff[ni] = ((tdw[ni] - 1) * Cin + ci0 + 4 * ml) * 2;
+  }
+
+  // per-lane opix streams s=0,1: opix = kt*BK + g*8 + s*4 + jl
+  int wo_[2], hrel_[2], habs_[2], op_[2], wob_[2];
+  int m5 = 0;  // (R0 mod WROWS); R0 = flat output row of opix0
+  int wo0, hoabs0, R0;
+  {
+    int opix0 = kt0 * BK;
+    wo0 = opix0 % Wo;
+    R0 = opix0 / Wo;
+    hoabs0 = R0 % Ho;
+    m5 = R0 % WROWS;
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      int off = g * 8 + s * 4 + jl;
+      int o = opix0 + off;
+      op_[s] = o;
+      int w = wo0 + off;
+      int hr = 0, ha = hoabs0;
+      while (w >= Wo) {
+        w -= Wo;
+        ++hr;
+        if (++ha == Ho) ha = 0;
+      }
+      wo_[s] = w;
+      hrel_[s] = hr;
+      habs_[s] = ha;
+      wob_[s] = w * Cin * 2;
+    }
+  }
+  int Floaded = R0 - 2;
+
+  auto advance = [&]() {
+    // uniform trackers
+    wo0 += BK;
+    while (wo0 >= Wo) {
+      wo0 -= Wo;
+      ++R0;
+      if (++hoabs0 == Ho) hoabs0 = 0;
+      if (++m5 == WROWS) m5 = 0;
+    }
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      op_[s] += BK;
+      wo_[s] += BK;
+      while (wo_[s] >= Wo) {
+        wo_[s] -= Wo;
+        ++hrel_[s];
+        if (++habs_[s] == Ho) habs_[s] = 0;
+      }
+      wob_[s] = wo_[s] * Cin * 2;
+    }
+    // hrel_ is relative to the OLD R0 after uniform advance moved it;
+    // re-anchor: hrel_new = (old hrel + old R0) - new R0 handled by
+    // tracking against R0 directly below (hrel_ -= deltaR0 done in
+    // caller via ranchor)
+  };
+
+  auto issue_A = [&](int buf) {
+    // v6-style blocked A deposit (one instr per wave)
+    int cs6 = (lane >> 3) & 3;
+    int h6 = lane & 1;
+    int q6 = (lane >> 5) & 1;
+    int r26 = (lane & 7) >> 1;
+    int opoff = q6 * 4 + r26;
+    int opix = (op_[0] - (g * 8 + jl)) + wid * 8 + opoff;
+    // (op_[0] - lane B-offset) = opix0 of this tile
+    int co8 = cs6 * 2 + h6;
+    const char* src;
+    if (opix < Mi) {
+      src = reinterpret_cast<const char*>(
+          gy + (int64_t)opix * Cout + tile_m + co8 * 8);
+    } else {
+      src = reinterpret_cast<const char*>(zero_page) + co8 * 16;
+    }
+    __builtin_amdgcn_global_load_lds(
+        (const __attribute__((address_space(1))) void*)src,
+        (__attribute__((address_space(3))) void*)(
+            smem + RAW_A0 + buf * 4096 + wid * 1024),
+        16, 0, 0);
+  };
+
+  auto load_rows = [&](int target) {
+    for (int F = Floaded + 1; F <= target; ++F) {
+      int Fc = F < 0 ? 0 : (F >= Frows ? Frows - 1 : F);
+      int slot = F % WROWS;
+      if (slot < 0) slot += WROWS;
+      const char* base =
+          reinterpret_cast<const char*>(x) + (int64_t)Fc * rowpitch;
+      char* dst = smem + WIN + slot * rowpitch;
+      int nq = rowpitch >> 10;
+      for (int q = wid; q < nq; q += 4) {
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) void*)(
+                base + q * 1024 + lane * 16),
+            (__attribute__((address_space(3))) void*)(dst + q * 1024),
+            16, 0, 0);
+      }
+    }
+    Floaded = target;
+  };
+
+  f32x4 acc[MREP][NREP] = {};
+  int cur = 0;
+  if (kt0 < kt1) {
+    issue_A(0);
+    int R1 = R0 + (wo0 + BK - 1) / Wo;
+    load_rows(R1 + 1);
+    __syncthreads();
+  }
+  for (int t = kt0; t < kt1; ++t) {
+    // --- snapshot tile-t fragment address bases ------------------------
+    // per s, per dh: window row byte (ring slot) + validity
+    int rbase[2][3];
+    bool rok[2][3];
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+#pragma unroll
+      for (int dh = 0; dh < 3; ++dh) {
+        int hi = habs_[s] + dh - 1;
+        rok[s][dh] = (unsigned)hi < (unsigned)Hi && op_[s] < Mi;
+        int r5 = m5 + hrel_[s] + dh - 1;
+        while (r5 >= WROWS) r5 -= WROWS;
+        if (r5 < 0) r5 += WROWS;
+        rbase[s][dh] = WIN + r5 * rowpitch;
+      }
+    }
+    int wob_s[2] = {wob_[0], wob_[1]};
+    int wo_s[2] = {wo_[0], wo_[1]};
+
+    // --- issue tile t+1 loads (fly under the MFMAs) --------------------
+    if (t + 1 < kt1) {
+      int oldR0 = R0;
+      advance();
+      int dR = R0 - oldR0;
+      hrel_[0] -= dR;
+      hrel_[1] -= dR;
+      issue_A(cur ^ 1);
+      int R1 = R0 + (wo0 + BK - 1) / Wo;
+      load_rows(R1 + 1);
+    }
+
+    // --- MFMA: A from blocked raw (tr), B straight from the window -----
+    {
+      char* rawA = smem + RAW_A0 + cur * 4096;
+      TrPair pa[MREP];
+#pragma unroll
+      for (int mi = 0; mi < MREP; ++mi)
+        tr_issue((unsigned)(size_t)(
+                     rawA + ((2 * g) * 4 + mi) * 128 + l15 * 8),
+                 pa[mi]);
+      bf16x8 a_frag[MREP];
+#pragma unroll
+      for (int mi = 0; mi < MREP; ++mi) a_frag[mi] = tr_wait<0>(pa[mi]);
+#pragma unroll
+      for (int ni = 0; ni < NREP; ++ni) {
+        // per-lane run addresses for the two k-quads (s = 0, 1)
+        unsigned ab[2];
+#pragma unroll
+        for (int s = 0; s < 2; ++s) {
+          int dh = tdh[ni];
+          bool ok = nok[ni] && rok[s][dh] &&
+                    (unsigned)(wo_s[s] + tdw[ni] - 1) < (unsigned)Wi;
+          int a = rbase[s][dh] + wob_s[s] + noff[ni];
+          ab[s] = ok ? (unsigned)a : (unsigned)(ZBLK + 8 * ml);
+        }
+        unsigned base32 = (unsigned)(size_t)smem;
+        unsigned long long q0, q1;
+        tr_issue1(base32 + ab[0], q0);
+        tr_issue1(base32 + ab[1], q1);
+        union {
+          unsigned long long u[2];
+          bf16x8 v;
+        } cb;
+        cb.u[0] = tr_wait1<1>(q0);
+        cb.u[1] = tr_wait1<0>(q1);
+        bf16x8 b_frag = cb.v;
+#pragma unroll
+        for (int mi = 0; mi < MREP; ++mi)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag, acc[mi][ni], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < MREP; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < NREP; ++ni) {
+      int col = tile_n + wid * WN + ni * 16 + (lane & 15);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = tile_m + mi * 16 + (lane >> 4) * 4 + j;
+        partial[(int64_t)row * Np + col] = acc[mi][ni][j];
+      }
+    }
+  }
+}
+
+namespace conv_wrw6 {
+constexpr int BM = 64, BN = 576, BK = 32;
+constexpr int WN = 144;
+constexpr int MREP = 4, NREP = 9;
+constexpr int RAW_A0 = 0;            //  4 KiB
+constexpr int RAW_B0 = 4 * 1024;     // 36 KiB
+constexpr int BUF_STRIDE = 40 * 1024;
+}  // namespace conv_wrw6
 
 __global__ __launch_bounds__(256) void conv_wrw6_kernel(
     const __hip_bfloat16* __restrict__ gy,  // (M, Cout) row-major
@@ -1674,6 +1961,59 @@ at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,
     force_v5 = (e && e[0] == '5') ? 1 : 0;
     force_v6 = (e && e[0] == '6') ? 1 : 0;
   }
+  static int force_v7 = -1;
+  if (force_v7 < 0) {
+    const char* e = getenv("TURBOPRUNE_WRW");
+    force_v7 = (e && e[0] == '7') ? 1 : 0;
+  }
+  bool v7_ok = KH == 3 && KW == 3 && stride == 1 && pad == 1 &&
+               Hi == Wi && Ho == Wo && Hi == Ho && Wo >= 28 &&
+               (Wi * Cin * 2) % 1024 == 0 &&
+               Wi * Cin * 2 <= conv_wrw7::MAXROW;
+  if ((force_v7 || (!use_db && !use_v1 && !force_v2 && !force_v3 &&
+                    !force_v4 && !force_v5 && !force_v6)) &&
+      v7_ok) {
+    // v7: ring window + direct tr_b16 window fragments
+    constexpr int BM7 = conv_wrw7::BM, BN7 = conv_wrw7::BN,
+                  BK7 = conv_wrw7::BK;
+    int gm = (Cout + BM7 - 1) / BM7;
+    int gn = (K + BN7 - 1) / BN7;
+    int Mp7 = gm * BM7;
+    int Np7 = gn * BN7;
+    int tiles7 = gm * gn;
+    int total_kt = (int)((M + BK7 - 1) / BK7);
+    int splitk = 1;
+    while (tiles7 * splitk < 1024 && splitk * 2 <= total_kt &&
+           splitk < 512)
+      splitk *= 2;
+    auto partial = at::empty({splitk, (int64_t)Mp7, (int64_t)Np7},
+                             x.options().dtype(at::kFloat));
+    static at::Tensor zp7;
+    if (!zp7.defined() || zp7.device() != x.device())
+      zp7 = at::zeros({64}, x.options());
+    constexpr int kLds7 = conv_wrw7::LDS_BYTES;
+    static bool attr7 = false;
+    if (!attr7) {
+      (void)hipFuncSetAttribute(
+          reinterpret_cast<const void*>(conv_wrw7_kernel),
+          hipFuncAttributeMaxDynamicSharedMemorySize, kLds7);
+      attr7 = true;
+    }
+    hipLaunchKernelGGL(conv_wrw7_kernel, dim3(tiles7, splitk), dim3(256),
+                       kLds7, stream,
+                       reinterpret_cast<const __hip_bfloat16*>(gy.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(x.data_ptr()),
+                       reinterpret_cast<const __hip_bfloat16*>(
+                           zp7.data_ptr()),
+                       partial.data_ptr<float>(), Nb, Hi, Cin, Cout,
+                       Mp7, Np7, gn);
+    auto gw = at::empty({Cout, Cin, KH, KW},
+                        gy.options().dtype(at::kFloat).memory_format(
+                            at::MemoryFormat::ChannelsLast));
+    launch_wrw_reduce(partial, gw, Mp7, Np7, splitk, K, Cout, stream);
+    return gw;
+  }
+
   if ((force_v6 ||
        (!use_db && !use_v1 && !force_v2 && !force_v3 && !force_v4 &&
         !force_v5)) &&
