@@ -215,7 +215,7 @@ def _load_table() -> None:
 # kernel change).
 _CONV_TABLE: dict = {
     (64, 64, 1, 1): {"fwd": False, "gradin": False, "wrw": False},
-    (64, 64, 3, 1): {"fwd": False, "gradin": False, "wrw": False},
+    (64, 64, 3, 1): {"fwd": False, "gradin": False, "wrw": True},
     (64, 256, 1, 1): {"fwd": True, "gradin": False, "wrw": False},
     (128, 128, 3, 1): {"fwd": False, "gradin": False, "wrw": False},
     (128, 128, 3, 2): {"fwd": True, "gradin": False, "wrw": False},
