@@ -1,17 +1,28 @@
-// Implicit-GEMM conv WEIGHT gradient (wrw) on MFMA — experimental.
+// Implicit-GEMM conv WEIGHT gradient (wrw) on MFMA.
 //
 //   gw[co][dh,dw,ci] = sum_opix gy[opix][co] * x[n, ho*s-p+dh, wo*s-p+dw, ci]
 //
 // A TN GEMM: both operands are M-major over output pixels (the deep
-// contraction, N*Ho*Wo up to ~1.6M), so tiles are loaded 16B-coalesced
-// along their fast dims (co / ci) and TRANSPOSED into K-major LDS images
-// by ds_write scatter (glds cannot transpose — guide §5.4 rule 21).
-// Split-K over opix chunks (gridDim.y slabs -> fp32 partials -> the
-// shared splitk reduce), since the output (Cout x 9Cin) is tiny.
+// contraction, N*Ho*Wo up to ~1.6M) — the transposed-staging problem
+// that drove a seven-generation engineering campaign, fully measured
+// in profiles/r02_summary.md:
 //
-// Tile: C 128(co) x 128(tapci), BK = 64 output pixels. LDS images
-// [row][k] like gemm_bt (XOR-swizzled 16B blocks), so the MFMA fragment
-// reads are identical to gemm_bt's.
+//   v1  128x128x64 register staging, ds_write_b16 scatter (shipped r01)
+//   v2  BK=128, b128-grouped register transpose, 1024-workgroup splitK
+//   v3  wide-N 64x576 tiles (operands read ~once), 32-bit coords
+//   v4  global_load_lds raw staging + in-LDS transpose
+//   v5  persistent ring window of x rows (opt-in, TURBOPRUNE_WRW=5)
+//   two-stage slab-parallel splitK reduce (hidden 450 us at splitk=512)
+//   v6  ds_read_b64_tr_b16 hardware-transpose fragments straight from
+//       the glds-blocked raw image (DEFAULT for 3x3): no transpose
+//       pass, double-buffered raw, one barrier/tile, pipelined counted
+//       lgkm waits, incremental 32-bit corner offsets. Layer1
+//       (64,64,3,1): 309 us vs MIOpen 477 — dispatched native.
+//   v7  ring window + window-direct tr fragments (opt-in, =7)
+//
+// All generations remain selectable via TURBOPRUNE_WRW for A/B.
+// Split-K over opix chunks (gridDim.y slabs -> fp32 partials -> the
+// two-stage reduce), since the output (Cout x 9Cin) is tiny.
 
 #include <ATen/ATen.h>
 #include <ATen/hip/HIPContext.h>
