@@ -1928,6 +1928,201 @@ static void launch_wrw_reduce(const at::Tensor& partial, at::Tensor& gw,
   }
 }
 
+
+// ---- transpose-free TN GEMM: C[N,K] = sum_m A[m,N] * B[m,K] ------------
+//
+// The deep-K backward-weight GEMM (linear_bwd grad_w, 1x1-conv wrw):
+// both operands are m-major, so the classic path materializes TWO
+// transposes first (transpose2d + gemm_bt — ~2 extra full passes).
+// Here both tiles stage ROW-MAJOR via glds in tr_b16-blocked form
+// ([4 m][16 col] 128 B blocks — the per-lane source choice builds the
+// blocking for free) and the MFMA fragments come from
+// ds_read_b64_tr_b16 (semantics probed in scripts/probe_tr16.hip).
+// 128x128 C tiles (2x2 waves), BKM=32 m per iteration, double-buffered
+// 32 KiB LDS, split-K with the two-stage slab reduce.
+namespace gemm_tn {
+constexpr int BN = 128, BKC = 128, BKM = 32;
+constexpr int WARPS = 2;  // 2x2
+constexpr int WT = 64;    // per-wave C tile side
+constexpr int REP = 4;
+constexpr int TILE_B = BKM * BN * 2;  // 8 KiB per operand tile
+}  // namespace gemm_tn
+
+__global__ __launch_bounds__(256) void gemm_tn_kernel(
+    const __hip_bfloat16* __restrict__ A,  // (M, N) row-major
+    const __hip_bfloat16* __restrict__ B,  // (M, K) row-major
+    const __hip_bfloat16* __restrict__ zero_page,
+    float* __restrict__ partial,           // (slabs, Np, Kp)
+    int64_t M, int N, int K, int Np, int Kp, int grid_k) {
+  using namespace gemm_tn;
+  __shared__ char smem[2 * 2 * TILE_B];  // [buf][A|B]
+  auto sA = [&](int b) { return smem + b * 2 * TILE_B; };
+  auto sB = [&](int b) { return smem + b * 2 * TILE_B + TILE_B; };
+
+  int wg = blockIdx.x;
+  int n0 = (wg / grid_k) * BN;
+  int k0 = (wg % grid_k) * BKC;
+
+  int lane = threadIdx.x & (kWave - 1);
+  int wid = threadIdx.x / kWave;
+  int wr = wid >> 1, wc = wid & 1;
+  int g = lane >> 4;
+  int l15 = lane & 15;
+
+  int total_mt = (int)((M + BKM - 1) / BKM);
+  int per = (total_mt + gridDim.y - 1) / gridDim.y;
+  int mt0 = blockIdx.y * per;
+  int mt1 = min(mt0 + per, total_mt);
+  partial += (int64_t)blockIdx.y * Np * Kp;
+
+  // staging: per wave 4 instrs (2 per operand): instr covers m-quad
+  // (wid half? distribute 8 A + 8 B instrs over 4 waves: wave w does
+  // A instrs {w, w+4} ... simpler: wave w stages A quads {2w, 2w+1}
+  // via 2 instrs and B the same.
+  // per-lane: ns = l>>3 (col subtile), r = (l&7)>>1, h = l&1
+  int ns_ = lane >> 3;        // 0..7
+  int r_ = (lane & 7) >> 1;   // m row within quad
+  int h_ = lane & 1;          // 8-col half of the 16-col subtile
+
+  auto stage = [&](int buf, int mt) {
+    int64_t m_base = (int64_t)mt * BKM;
+#pragma unroll
+    for (int ii = 0; ii < 2; ++ii) {
+      int mq = wid * 2 + ii;          // m quad 0..7
+      int64_t m = m_base + mq * 4 + r_;
+      int ca = n0 + ns_ * 16 + h_ * 8;
+      const char* srcA =
+          (m < M && ca + 8 <= N)
+              ? reinterpret_cast<const char*>(A + m * N + ca)
+              : reinterpret_cast<const char*>(zero_page) +
+                    (ns_ & 7) * 16;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)srcA,
+          (__attribute__((address_space(3))) void*)(sA(buf) + mq * 1024),
+          16, 0, 0);
+      int cb = k0 + ns_ * 16 + h_ * 8;
+      const char* srcB =
+          (m < M && cb + 8 <= K)
+              ? reinterpret_cast<const char*>(B + m * K + cb)
+              : reinterpret_cast<const char*>(zero_page) +
+                    (ns_ & 7) * 16;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)srcB,
+          (__attribute__((address_space(3))) void*)(sB(buf) + mq * 1024),
+          16, 0, 0);
+    }
+  };
+
+  f32x4 acc[REP][REP] = {};
+  int cur = 0;
+  if (mt0 < mt1) {
+    stage(0, mt0);
+    __syncthreads();
+  }
+  for (int t = mt0; t < mt1; ++t) {
+    if (t + 1 < mt1) stage(cur ^ 1, t + 1);
+    {
+      // fragments: [4 m][16 col] blocks at (mq*8 + ns)*128;
+      // wave row range: wr*64 + mi*16; col range: wc*64 + ni*16
+      // NOTE the k-quad pair (h2 = 0,1 -> mq = 2g, 2g+1) sits 1024 B
+      // apart in this 8-subtile layout, so paired tr_issue (+512)
+      // does not apply — single reads with explicit offsets.
+      unsigned long long qa[REP][2];
+#pragma unroll
+      for (int mi = 0; mi < REP; ++mi) {
+        int ns = (wr * 64 + mi * 16) >> 4;
+        unsigned a0 = (unsigned)(size_t)(
+            sA(cur) + ((2 * g) * 8 + ns) * 128 + l15 * 8);
+        tr_issue1(a0, qa[mi][0]);
+        tr_issue1(a0 + 1024, qa[mi][1]);
+      }
+      bf16x8 a_frag[REP];
+#pragma unroll
+      for (int mi = 0; mi < REP; ++mi) {
+        union {
+          unsigned long long u[2];
+          bf16x8 v;
+        } ca;
+        ca.u[0] = tr_wait1<0>(qa[mi][0]);
+        ca.u[1] = tr_wait1<0>(qa[mi][1]);
+        a_frag[mi] = ca.v;
+      }
+#pragma unroll
+      for (int ni = 0; ni < REP; ++ni) {
+        int ns = (wc * 64 + ni * 16) >> 4;
+        unsigned b0a = (unsigned)(size_t)(
+            sB(cur) + ((2 * g) * 8 + ns) * 128 + l15 * 8);
+        unsigned long long qb0, qb1;
+        tr_issue1(b0a, qb0);
+        tr_issue1(b0a + 1024, qb1);
+        union {
+          unsigned long long u[2];
+          bf16x8 v;
+        } cb;
+        cb.u[0] = tr_wait1<1>(qb0);
+        cb.u[1] = tr_wait1<0>(qb1);
+        bf16x8 b_frag = cb.v;
+#pragma unroll
+        for (int mi = 0; mi < REP; ++mi)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a_frag[mi], b_frag, acc[mi][ni], 0, 0, 0);
+      }
+    }
+    __syncthreads();
+    cur ^= 1;
+  }
+
+#pragma unroll
+  for (int mi = 0; mi < REP; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < REP; ++ni) {
+      int col = k0 + wc * WT + ni * 16 + (lane & 15);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        int row = n0 + wr * WT + mi * 16 + (lane >> 4) * 4 + j;
+        partial[(int64_t)row * Kp + col] = acc[mi][ni][j];
+      }
+    }
+  }
+}
+
+// C[N,K] = A[M,N]^T B[M,K], bf16 in, bf16 out (fp32 accumulate).
+at::Tensor gemm_tn_bf16(const at::Tensor& A_in, const at::Tensor& B_in) {
+  using namespace gemm_tn;
+  auto A = A_in.contiguous();
+  auto B = B_in.contiguous();
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16 &&
+              B.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(A.dim() == 2 && B.dim() == 2 && A.size(0) == B.size(0));
+  TORCH_CHECK(A.size(1) % 8 == 0 && B.size(1) % 8 == 0,
+              "gemm_tn: N, K must be octet-aligned");
+  int64_t M = A.size(0);
+  int N = (int)A.size(1), K = (int)B.size(1);
+  int gn = (N + BN - 1) / BN, gk = (K + BKC - 1) / BKC;
+  int Np = gn * BN, Kp = gk * BKC;
+  int tiles = gn * gk;
+  int total_mt = (int)((M + BKM - 1) / BKM);
+  int splitk = 1;
+  while (tiles * splitk < 1024 && splitk * 2 <= total_mt && splitk < 512)
+    splitk *= 2;
+  auto stream = at::hip::getCurrentHIPStream();
+  auto partial = at::empty({splitk, (int64_t)Np, (int64_t)Kp},
+                           A.options().dtype(at::kFloat));
+  static at::Tensor zp;
+  if (!zp.defined() || zp.device() != A.device())
+    zp = at::zeros({64}, A.options());
+  hipLaunchKernelGGL(gemm_tn_kernel, dim3(tiles, splitk), dim3(256), 0,
+                     stream,
+                     reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(B.data_ptr()),
+                     reinterpret_cast<const __hip_bfloat16*>(zp.data_ptr()),
+                     partial.data_ptr<float>(), M, N, K, Np, Kp, gk);
+  // reduce slabs -> bf16 C
+  auto Cf = at::empty({N, K}, A.options().dtype(at::kFloat));
+  launch_wrw_reduce(partial, Cf, Np, Kp, splitk, K, N, stream);
+  return Cf.to(at::kBFloat16);
+}
+
 // gy: (N, Cout, Ho, Wo) channels_last; x: (N, Cin, Hi, Wi) channels_last.
 // Returns grad_weight (Cout, Cin, KH, KW) channels_last bf16.
 at::Tensor conv2d_implicit_wrw(const at::Tensor& gy_in,

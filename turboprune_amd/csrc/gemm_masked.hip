@@ -30,6 +30,8 @@
 
 namespace turboprune {
 
+at::Tensor gemm_tn_bf16(const at::Tensor&, const at::Tensor&);  // conv_wrw.hip
+
 at::Tensor transpose2d(const at::Tensor&);  // transpose.hip
 
 using bf16x8 = __attribute__((ext_vector_type(8))) short;
@@ -364,9 +366,20 @@ std::tuple<at::Tensor, at::Tensor> linear_bwd(const at::Tensor& grad_y,
   // grad_x (M,K) = gy (M,N) @ w (N,K): B_t = w^T (K,N) K-major in N
   auto wT = transpose2d(w.contiguous());
   auto gx = gemm_route(gy2, wT, c10::nullopt, false);
-  // grad_w (N,K) = gy^T (N,M) @ x (M,K): A = gy^T, B_t = x^T (K,M)
-  auto gw = gemm_route(transpose2d(gy2), transpose2d(x2), c10::nullopt,
-                       false);
+  // grad_w (N,K) = gy^T (N,M) @ x (M,K): the transpose-free TN GEMM
+  // (tr_b16 fragments, conv_wrw.hip) replaces the former
+  // transpose2d+gemm_bt composition — measured 1.3-1.8x faster on
+  // every DeiT training shape (r2t); TURBOPRUNE_TN=0 restores the old
+  // path for A/B.
+  static int use_tn = -1;
+  if (use_tn < 0) {
+    const char* e = getenv("TURBOPRUNE_TN");
+    use_tn = (e && e[0] == '0') ? 0 : 1;
+  }
+  auto gw = use_tn
+                ? gemm_tn_bf16(gy2, x2)
+                : gemm_route(transpose2d(gy2), transpose2d(x2),
+                             c10::nullopt, false);
   auto x_sizes = x.sizes().vec();
   return {gx.reshape(x_sizes), gw};
 }
