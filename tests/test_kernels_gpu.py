@@ -447,3 +447,21 @@ def test_synflow_linearize_restore(ext):
     assert torch.equal(t, orig.abs())
     TF.synflow_restore_(t, sign)
     assert torch.equal(t, orig)
+
+
+@pytest.mark.parametrize("mnk", [
+    (50432, 1152, 384),   # DeiT qkv grad_w
+    (4096, 384, 1536),
+    (1000, 64, 64),       # small M tail
+    (50176, 256, 1024),
+])
+def test_gemm_tn_matches_torch(ext, mnk):
+    """Transpose-free TN GEMM (tr_b16 fragments): C = A^T B."""
+    M, N, K = mnk
+    torch.manual_seed(N + K)
+    A = (torch.rand(M, N, device=DEV) - 0.5).to(torch.bfloat16)
+    B = (torch.rand(M, K, device=DEV) - 0.5).to(torch.bfloat16)
+    got = ext.gemm_tn_bf16(A, B)
+    ref = A.t().float() @ B.float()
+    err = (got.float() - ref).abs().max().item()
+    assert err < 0.02 * max(ref.abs().max().item(), 1.0), err
