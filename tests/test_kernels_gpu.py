@@ -465,3 +465,13 @@ def test_gemm_tn_matches_torch(ext, mnk):
     ref = A.t().float() @ B.float()
     err = (got.float() - ref).abs().max().item()
     assert err < 0.02 * max(ref.abs().max().item(), 1.0), err
+
+
+def test_colsum_bf16_matches_torch(ext):
+    torch.manual_seed(7)
+    for (m, n) in [(50432, 1152), (1000, 8), (4097, 384)]:
+        A = (torch.rand(m, n, device=DEV) - 0.5).to(torch.bfloat16)
+        got = ext.colsum_bf16(A)
+        ref = A.float().sum(0)
+        assert torch.allclose(got, ref, atol=0.5, rtol=1e-3), \
+            (m, n, (got - ref).abs().max().item())

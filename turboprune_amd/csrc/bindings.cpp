@@ -8,6 +8,7 @@ at::Tensor masked_abs_score(const at::Tensor&, const at::Tensor&,
                             const at::Tensor&);
 void bernoulli_mask_(at::Tensor, double, int64_t);
 at::Tensor sign_abs_(at::Tensor);
+at::Tensor colsum_bf16(const at::Tensor&);
 at::Tensor gemm_tn_bf16(const at::Tensor&, const at::Tensor&);
 void mul_sign_(at::Tensor, const at::Tensor&);
 void sgd_step_(at::Tensor, const at::Tensor&, at::Tensor, const at::Tensor&,
@@ -96,6 +97,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_tn_bf16", &turboprune::gemm_tn_bf16,
         "C[N,K] = A[M,N]^T B[M,K] — transpose-free deep-K TN GEMM "
         "(tr_b16 fragments)");
+  m.def("colsum_bf16", &turboprune::colsum_bf16,
+        "fp32 column sums of a bf16 matrix (bias gradients)");
   m.def("sign_abs_", &turboprune::sign_abs_,
         "SynFlow linearize: t = |t| in place, returns int8 signs (K10)");
   m.def("mul_sign_", &turboprune::mul_sign_,

@@ -8,6 +8,8 @@
 #include <ATen/ATen.h>
 #include <ATen/hip/HIPContext.h>
 
+#include <algorithm>
+
 #include "common.h"
 
 namespace turboprune {
@@ -302,6 +304,68 @@ void colsum2_atomic(const at::Tensor& pa, const at::Tensor& pb,
                      dim3(kBlock), 0, stream, pa.data_ptr<float>(),
                      pb.data_ptr<float>(), oa.data_ptr<float>(),
                      ob.data_ptr<float>(), R, C);
+}
+
+// ---- bf16 column sum (bias gradients) -----------------------------------
+// grad_b = sum_m gy[m, n]: one streaming pass (the eager torch reduce
+// was 5.5% of the DeiT step, r2w). bn_reduce-style grid: 8 column
+// octets x 32 row lanes per block, LDS tree, one atomicAdd per column
+// per block (fp32 atomics — same determinism class as the bn dgamma
+// path).
+__global__ void colsum_bf16_kernel(const __hip_bfloat16* __restrict__ A,
+                                   float* __restrict__ out, int64_t rows,
+                                   int C) {
+  __shared__ float ls[8][8][32];
+  int oct_in_blk = threadIdx.x & 7;
+  int lane = threadIdx.x >> 3;
+  int oct = blockIdx.y * 8 + oct_in_blk;
+  int c0 = oct * 8;
+  float s[8] = {};
+  if (c0 < C) {
+    for (int64_t r = (int64_t)blockIdx.x * 32 + lane; r < rows;
+         r += (int64_t)gridDim.x * 32) {
+      uint4 v = *reinterpret_cast<const uint4*>(A + r * C + c0);
+      const __hip_bfloat16* p =
+          reinterpret_cast<const __hip_bfloat16*>(&v);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) s[j] += __bfloat162float(p[j]);
+    }
+  }
+#pragma unroll
+  for (int j = 0; j < 8; ++j) ls[oct_in_blk][j][lane] = s[j];
+  __syncthreads();
+  for (int step = 16; step > 0; step >>= 1) {
+    if (lane < step) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        ls[oct_in_blk][j][lane] += ls[oct_in_blk][j][lane + step];
+    }
+    __syncthreads();
+  }
+  if (lane == 0 && c0 < C) {
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      atomicAdd(&out[c0 + j], ls[oct_in_blk][j][0]);
+  }
+}
+
+at::Tensor colsum_bf16(const at::Tensor& A_in) {
+  auto A = A_in.contiguous();
+  TORCH_CHECK(A.is_cuda() && A.scalar_type() == at::kBFloat16 &&
+              A.dim() == 2 && A.size(1) % 8 == 0);
+  int64_t rows = A.size(0);
+  int C = (int)A.size(1);
+  auto out = at::zeros({C}, A.options().dtype(at::kFloat));
+  int cb = (C / 8 + 7) / 8;
+  int64_t rb_want = (rows + 32 * 16 - 1) / (32 * 16);
+  int rb = (int)std::min<int64_t>(std::max<int64_t>(rb_want, 1),
+                                  std::max<int64_t>(4096 / cb, 8));
+  auto stream = at::hip::getCurrentHIPStream();
+  hipLaunchKernelGGL(colsum_bf16_kernel, dim3(rb, cb), dim3(256), 0,
+                     stream,
+                     reinterpret_cast<const __hip_bfloat16*>(A.data_ptr()),
+                     out.data_ptr<float>(), rows, C);
+  return out;
 }
 
 // ---- SynFlow linearize / restore ---------------------------

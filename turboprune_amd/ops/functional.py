@@ -125,11 +125,17 @@ class _MaskedLinearGemm(torch.autograd.Function):
     def backward(ctx, grad_y):
         x, w = ctx.saved_tensors
         ext = _backend.extension()
-        grad_x, grad_w = ext.linear_bwd(grad_y.contiguous(), x, w)
+        gy = grad_y.contiguous()
+        grad_x, grad_w = ext.linear_bwd(gy, x, w)
         grad_b = None
         if ctx.bias_dtype is not None:
-            grad_b = grad_y.reshape(-1, grad_y.shape[-1]).sum(0) \
-                .to(ctx.bias_dtype)
+            gy2 = gy.reshape(-1, gy.shape[-1])
+            if gy2.dtype == torch.bfloat16 and gy2.shape[-1] % 8 == 0:
+                # one streaming HIP pass (the eager reduce was 5.5% of
+                # the DeiT step, profiles/r02_bench_final.md)
+                grad_b = ext.colsum_bf16(gy2).to(ctx.bias_dtype)
+            else:
+                grad_b = gy2.sum(0).to(ctx.bias_dtype)
         return grad_x, grad_w, grad_b
 
 
