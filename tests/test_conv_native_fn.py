@@ -93,3 +93,41 @@ def test_resnet50_dispatch_envelope():
             (inside if ok else outside).append(name)
     assert outside == ["conv1"], outside
     assert len(inside) == 52, len(inside)
+
+
+def test_auto_backend_table_logic():
+    """Dispatch-table plumbing (no GPU needed): measured winners route
+    per (shape, op); unknown shapes use the declared defaults; the
+    Function is skipped entirely for all-library shapes."""
+    from turboprune_amd.ops.conv_native import AutoBackend, _load_table
+    _load_table()
+    t = AutoBackend.table
+    assert len(t) >= 20
+    # measured winners (profiles/r02_conv_dispatch.md + r2r)
+    assert t[(64, 256, 1, 1)]["fwd"] is True
+    assert t[(64, 64, 3, 1)]["wrw"] is True      # tr_b16 v6 beat MIOpen
+    assert t[(64, 64, 3, 1)]["fwd"] is False
+    assert t[(512, 512, 3, 1)]["wrw"] is False   # MIOpen keeps it
+    # any_native gates the Function entry
+    assert AutoBackend.any_native(256, 64, 1, 1)       # fwd native
+    assert AutoBackend.any_native(64, 64, 3, 1)        # wrw native
+    assert not AutoBackend.any_native(64, 64, 1, 1)    # all library
+    # unknown shape: defaults
+    d = AutoBackend._ops_for((9999, 9999, 3, 1))
+    assert d == AutoBackend.DEFAULT_NATIVE
+
+
+def test_auto_table_json_override(tmp_path, monkeypatch):
+    import json
+    p = tmp_path / "table.json"
+    p.write_text(json.dumps({"[1, 2, 3, 4]": {"fwd": True,
+                                              "gradin": False,
+                                              "wrw": False}}))
+    monkeypatch.setenv("TURBOPRUNE_CONV_TABLE", str(p))
+    from turboprune_amd.ops.conv_native import AutoBackend, _load_table
+    try:
+        _load_table()
+        assert AutoBackend.table[(1, 2, 3, 4)]["fwd"] is True
+    finally:
+        monkeypatch.delenv("TURBOPRUNE_CONV_TABLE")
+        _load_table()  # restore the built-in table
