@@ -112,7 +112,7 @@ class _Prefetcher:
     """
 
     DEPTH = 2
-    GATHER_THREADS = 8
+    GATHER_THREADS = int(os.environ.get("TURBOPRUNE_GATHER_THREADS", "8"))
 
     def __init__(self, device: torch.device, batch_shape, work):
         from concurrent.futures import ThreadPoolExecutor
