@@ -189,6 +189,10 @@ class _Prefetcher:
             self._put(None)
         except BaseException as e:  # surfaced to the consumer
             self._put(e)
+        finally:
+            # a fresh prefetcher is created per epoch: release the
+            # gather pool's threads when this epoch's work ends
+            self._pool.shutdown(wait=False)
 
     def __iter__(self):
         try:
