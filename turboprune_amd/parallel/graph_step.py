@@ -103,6 +103,15 @@ class GraphedTrainStep:
         del last
         torch.cuda.synchronize()
 
+        # autograd must have accumulated IN PLACE into the flat views;
+        # a rebound .grad would make the captured all-reduce silently
+        # reduce stale memory
+        lo = self.flat.flat.data_ptr()
+        hi = lo + self.flat.flat.numel() * 4
+        for p in self.flat.params:
+            assert p.grad is not None and lo <= p.grad.data_ptr() < hi, \
+                "a grad escaped the flat buffer; cannot capture"
+
         self.graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(self.graph):
             self.loss = body()
