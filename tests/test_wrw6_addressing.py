@@ -128,3 +128,43 @@ def test_tr_semantics_match_probe():
     out = tr_read(lds, [l * 8 for l in range(16)])
     for i in range(16):
         assert out[i] == [i, i + 16, i + 32, i + 48]
+
+
+def simulate_gemm_tn_operand(which="B"):
+    """gemm_tn (csrc/conv_wrw.hip): stage instr (wave wid, ii) deposits
+    1 KiB at mq*1024 (mq = wid*2+ii); lane -> (ns = l>>3, r = (l&7)>>1,
+    h = l&1) loading col-octet ns*16 + h*8 of m-row mq*4 + r. Fragment:
+    block (mq*8 + ns)*128, k-halves at +0/+1024."""
+    def val(m, col):
+        return (which, m, col)
+
+    lds = {}
+    for mq in range(8):
+        base_elem = mq * 1024 // 2
+        for lane in range(64):
+            ns = lane >> 3
+            r = (lane & 7) >> 1
+            h = lane & 1
+            m = mq * 4 + r
+            c0 = ns * 16 + h * 8
+            for e in range(8):
+                lds[base_elem + lane * 8 + e] = val(m, c0 + e)
+
+    for g in range(4):          # lane group = k octet
+        for ns in range(8):     # 16-col subtile
+            frag = [[None] * 8 for _ in range(16)]
+            for h2 in range(2):
+                base = ((2 * g + h2) * 8 + ns) * 128
+                addrs = [base + l15 * 8 for l15 in range(16)]
+                out = tr_read(lds, addrs)
+                for i in range(16):
+                    for j in range(4):
+                        frag[i][h2 * 4 + j] = out[i][j]
+            for i in range(16):
+                for k in range(8):
+                    expect = val(g * 8 + k, ns * 16 + i)
+                    assert frag[i][k] == expect, (g, ns, i, k)
+
+
+def test_gemm_tn_operand_addressing():
+    simulate_gemm_tn_operand()
